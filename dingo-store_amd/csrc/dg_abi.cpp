@@ -1,0 +1,1140 @@
+// dg_abi.cpp — C-ABI implementation (include/dingo_gpu.h) of the
+// MI355X-native dingo-store vector-search path.  Host orchestration around
+// the kernels in kernels.hip.cpp; plain library GEMMs (query x centroid,
+// query x database) via rocBLAS.  No CPU compute fallback anywhere: without
+// a HIP device every entry point fails with DG_ENOGPU.
+#include <cmath>
+#include <cstdarg>
+#include <cstdio>
+#include <cstring>
+#include <algorithm>
+#include <vector>
+
+#include "dg_internal.h"
+
+static thread_local char g_err[1024] = "";
+
+void dg_set_error(const char* fmt, ...) {
+  va_list ap;
+  va_start(ap, fmt);
+  vsnprintf(g_err, sizeof(g_err), fmt, ap);
+  va_end(ap);
+}
+
+extern "C" void dg_last_error(char* buf, int64_t len) {
+  if (buf && len > 0) {
+    strncpy(buf, g_err, len - 1);
+    buf[len - 1] = 0;
+  }
+}
+
+extern "C" int dg_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+extern "C" const char* dg_build_info(void) {
+  return "dingo_gpu 0.1 gfx950 hip+rocblas fp32";
+}
+
+// ---------------- device buffer helpers ----------------
+static dg_status dbuf_reserve(dg_dbuf& b, size_t bytes, hipStream_t s,
+                              bool keep) {
+  if (bytes <= b.cap) {
+    b.bytes = bytes;
+    return DG_OK;
+  }
+  size_t newcap = std::max(bytes, b.cap + b.cap / 2);
+  void* np = nullptr;
+  if (hipMalloc(&np, newcap) != hipSuccess) {
+    dg_set_error("hipMalloc(%zu) failed", newcap);
+    return DG_ENOMEM;
+  }
+  if (keep && b.p && b.bytes) {
+    if (hipMemcpyAsync(np, b.p, b.bytes, hipMemcpyDeviceToDevice, s) !=
+        hipSuccess) {
+      (void)hipFree(np);
+      dg_set_error("grow copy failed");
+      return DG_EINTERNAL;
+    }
+    (void)hipStreamSynchronize(s);
+  }
+  if (b.p) (void)hipFree(b.p);
+  b.p = np;
+  b.cap = newcap;
+  b.bytes = bytes;
+  return DG_OK;
+}
+
+static void dbuf_free(dg_dbuf& b) {
+  if (b.p) (void)hipFree(b.p);
+  b = {};
+}
+
+// ---------------- mt19937 (train subsample/init determinism; standard
+// algorithm, independent implementation — mirrors faiss RandomGenerator
+// semantics restated in oracle.c) ----------------
+struct Mt19937 {
+  uint32_t mt[624];
+  int idx;
+  explicit Mt19937(uint32_t seed) {
+    mt[0] = seed;
+    for (int i = 1; i < 624; i++)
+      mt[i] = 1812433253u * (mt[i - 1] ^ (mt[i - 1] >> 30)) + i;
+    idx = 624;
+  }
+  uint32_t next() {
+    if (idx >= 624) {
+      for (int i = 0; i < 624; i++) {
+        uint32_t y = (mt[i] & 0x80000000u) | (mt[(i + 1) % 624] & 0x7fffffffu);
+        mt[i] = mt[(i + 397) % 624] ^ (y >> 1);
+        if (y & 1) mt[i] ^= 2567483615u;
+      }
+      idx = 0;
+    }
+    uint32_t y = mt[idx++];
+    y ^= y >> 11;
+    y ^= (y << 7) & 2636928640u;
+    y ^= (y << 15) & 4022730752u;
+    y ^= y >> 18;
+    return y;
+  }
+  int64_t rand_int(int64_t max) { return (int64_t)(next() % (uint64_t)max); }
+  float rand_float() { return next() * (1.0f / 4294967296.0f); }
+};
+
+static void rand_perm(std::vector<int64_t>& perm, int64_t n, uint32_t seed) {
+  perm.resize(n);
+  for (int64_t i = 0; i < n; i++) perm[i] = i;
+  Mt19937 rng(seed);
+  for (int64_t i = 0; i + 1 < n; i++)
+    std::swap(perm[i], perm[i + rng.rand_int(n - i)]);
+}
+
+// ---------------- device guard ----------------
+struct DeviceGuard {
+  int prev = -1;
+  explicit DeviceGuard(int dev) {
+    (void)hipGetDevice(&prev);
+    if (dev >= 0 && dev != prev) (void)hipSetDevice(dev);
+  }
+  ~DeviceGuard() {
+    if (prev >= 0) (void)hipSetDevice(prev);
+  }
+};
+
+// sgemm: dots[rows x cols] (row-major) = X[rows x d] * Y[cols x d]^T
+static dg_status sgemm_dots(dg_index* ix, const float* X, int64_t rows,
+                            const float* Y, int64_t cols, int32_t d,
+                            float* dots) {
+  const float one = 1.0f, zero = 0.0f;
+  DG_ROCBLAS_CHECK(rocblas_sgemm(
+      ix->blas, rocblas_operation_transpose, rocblas_operation_none,
+      (rocblas_int)cols, (rocblas_int)rows, (rocblas_int)d, &one, Y,
+      (rocblas_int)d, X, (rocblas_int)d, &zero, dots, (rocblas_int)cols));
+  return DG_OK;
+}
+
+// ---------------- create / destroy ----------------
+extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
+  if (!out || !dp) {
+    dg_set_error("null arg");
+    return DG_EINVAL;
+  }
+  dg_index_desc desc = *dp;
+  if (desc.d <= 0 || desc.d % 4 != 0 || desc.d > 8192) {
+    dg_set_error(
+        "dimension %d unsupported (need 0 < d <= 8192, d %% 4 == 0)", desc.d);
+    return desc.d <= 0 ? DG_EINVAL : DG_ENOT_SUPPORT;
+  }
+  if (desc.metric < 0 || desc.metric > 2) {
+    dg_set_error("bad metric %d", desc.metric);
+    return DG_EINVAL;
+  }
+  if (desc.kind == DG_INDEX_IVF_PQ) {
+    dg_set_error("IVF_PQ kernels land in a later round (SURVEY.md §7 item 6)");
+    return DG_ENOT_SUPPORT;
+  }
+  if (desc.kind != DG_INDEX_FLAT && desc.kind != DG_INDEX_IVF_FLAT) {
+    dg_set_error("bad kind %d", desc.kind);
+    return DG_EINVAL;
+  }
+  if (desc.kind == DG_INDEX_IVF_FLAT && desc.nlist <= 0)
+    desc.nlist = 2048;  // kCreateIvfFlatParamNcentroids, constant.h:177
+  if (dg_device_count() == 0) {
+    dg_set_error("no HIP device (the GPU path has no CPU fallback)");
+    return DG_ENOGPU;
+  }
+  dg_index* ix = new dg_index();
+  ix->desc = desc;
+  ix->device = desc.device >= 0 ? desc.device : 0;
+  DeviceGuard g(ix->device);
+  if (hipStreamCreate(&ix->stream) != hipSuccess ||
+      rocblas_create_handle(&ix->blas) != rocblas_status_success) {
+    dg_set_error("stream/rocblas init failed");
+    delete ix;
+    return DG_EINTERNAL;
+  }
+  rocblas_set_stream(ix->blas, ix->stream);
+  for (auto& e : ix->ev) (void)hipEventCreate(&e);
+  ix->events_ready = true;
+  if (desc.kind == DG_INDEX_FLAT) ix->trained = true;
+  *out = ix;
+  return DG_OK;
+}
+
+extern "C" void dg_index_destroy(dg_index* ix) {
+  if (!ix) return;
+  DeviceGuard g(ix->device);
+  (void)hipStreamSynchronize(ix->stream);
+  for (auto b :
+       {&ix->d_vectors, &ix->d_ids, &ix->d_assign, &ix->d_centroids,
+        &ix->d_cnorms, &ix->d_csr_offsets, &ix->d_csr_vectors, &ix->d_csr_ids,
+        &ix->d_csr_vnorms, &ix->d_list_mask, &ix->ws_queries, &ix->ws_qnorms,
+        &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
+        &ix->ws_small, &ix->ws_topk})
+    dbuf_free(*b);
+  for (auto& e : ix->ev)
+    if (e) (void)hipEventDestroy(e);
+  if (ix->blas) rocblas_destroy_handle(ix->blas);
+  if (ix->stream) (void)hipStreamDestroy(ix->stream);
+  delete ix;
+}
+
+// ---------------- centroids ----------------
+extern "C" dg_status dg_set_centroids(dg_index* ix, int32_t nlist,
+                                      const float* centroids) {
+  if (!ix || !centroids) return DG_EINVAL;
+  if (ix->desc.kind != DG_INDEX_IVF_FLAT) {
+    dg_set_error("set_centroids on non-IVF index");
+    return DG_EINVAL;
+  }
+  if (nlist <= 0 || nlist > ix->desc.nlist) {
+    dg_set_error("nlist %d out of range (create nlist %d)", nlist,
+                 ix->desc.nlist);
+    return DG_EINVAL;
+  }
+  std::unique_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  ix->desc.nlist = nlist;  // degrade semantics, ivf_flat.cc:676-680
+  size_t bytes = (size_t)nlist * ix->desc.d * 4;
+  dg_status st = dbuf_reserve(ix->d_centroids, bytes, ix->stream, false);
+  if (st != DG_OK) return st;
+  DG_HIP_CHECK(hipMemcpyAsync(ix->d_centroids.p, centroids, bytes,
+                              hipMemcpyHostToDevice, ix->stream));
+  st = dbuf_reserve(ix->d_cnorms, (size_t)nlist * 4, ix->stream, false);
+  if (st != DG_OK) return st;
+  dgk::row_norms(ix->stream, (const float*)ix->d_centroids.p, nlist,
+                 ix->desc.d, (float*)ix->d_cnorms.p);
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  ix->trained = true;
+  ix->csr_valid = false;
+  return DG_OK;
+}
+
+extern "C" dg_status dg_get_centroids(dg_index* ix, float* out) {
+  if (!ix || !out) return DG_EINVAL;
+  if (!ix->trained || ix->desc.kind != DG_INDEX_IVF_FLAT) {
+    dg_set_error("not trained");
+    return DG_ENOT_TRAINED;
+  }
+  std::shared_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  DG_HIP_CHECK(hipMemcpy(out, ix->d_centroids.p,
+                         (size_t)ix->desc.nlist * ix->desc.d * 4,
+                         hipMemcpyDeviceToHost));
+  return DG_OK;
+}
+
+// ---------------- train (GPU k-means; faiss Clustering semantics
+// restated — see oracle.c for the algorithm statement) ----------------
+static dg_status assign_rows(dg_index* ix, const float* d_x, int64_t n,
+                             int32_t* d_assign) {
+  // chunked: dots[chunk x nlist] = X * C^T ; argmin
+  const int32_t nlist = ix->desc.nlist;
+  const int32_t d = ix->desc.d;
+  const int64_t chunk = std::max<int64_t>(
+      1, std::min<int64_t>(n, (int64_t)(512ull << 20) / ((size_t)nlist * 4)));
+  dg_status st =
+      dbuf_reserve(ix->ws_dots, (size_t)chunk * nlist * 4, ix->stream, false);
+  if (st != DG_OK) return st;
+  for (int64_t s0 = 0; s0 < n; s0 += chunk) {
+    int64_t c = std::min(chunk, n - s0);
+    st = sgemm_dots(ix, d_x + (size_t)s0 * d, c,
+                    (const float*)ix->d_centroids.p, nlist, d,
+                    (float*)ix->ws_dots.p);
+    if (st != DG_OK) return st;
+    dgk::argmin_rows(ix->stream, (const float*)ix->ws_dots.p,
+                     (const float*)ix->d_cnorms.p, c, nlist, ix->desc.metric,
+                     d_assign + s0);
+  }
+  return DG_OK;
+}
+
+extern "C" dg_status dg_train(dg_index* ix, int64_t n, const float* x) {
+  if (!ix || !x || n <= 0) {
+    dg_set_error("bad train args");
+    return DG_EINVAL;
+  }
+  if (ix->desc.kind == DG_INDEX_FLAT) return DG_OK;  // Flat needs no train
+  std::unique_lock lk(ix->rw);
+  if (ix->trained) return DG_OK;  // no-op, ivf_flat.cc:669-671
+  DeviceGuard g(ix->device);
+  const int32_t d = ix->desc.d;
+  // degrade: data < nlist => nlist = 1 (ivf_flat.cc:676-680)
+  if (n < ix->desc.nlist) ix->desc.nlist = 1;
+  const int32_t nlist = ix->desc.nlist;
+  const uint32_t seed = 1234;  // faiss ClusteringParameters.seed
+  const int32_t niter = 25;
+
+  // subsample to 256*nlist on host (faiss max_points_per_centroid)
+  const int64_t maxp = (int64_t)256 * nlist;
+  std::vector<int64_t> sel;
+  int64_t nt = n;
+  const float* src = x;
+  std::vector<float> staged;
+  if (n > maxp) {
+    std::vector<int64_t> perm;
+    rand_perm(perm, n, seed);
+    nt = maxp;
+    staged.resize((size_t)nt * d);
+    for (int64_t i = 0; i < nt; i++)
+      memcpy(&staged[(size_t)i * d], x + (size_t)perm[i] * d, (size_t)d * 4);
+    src = staged.data();
+  }
+  // upload train set
+  dg_dbuf d_td{}, d_grouped{}, d_asg{}, d_perm{}, d_off{};
+  dg_status st = dbuf_reserve(d_td, (size_t)nt * d * 4, ix->stream, false);
+  if (st == DG_OK)
+    st = dbuf_reserve(d_grouped, (size_t)nt * d * 4, ix->stream, false);
+  if (st == DG_OK) st = dbuf_reserve(d_asg, (size_t)nt * 4, ix->stream, false);
+  if (st == DG_OK) st = dbuf_reserve(d_perm, (size_t)nt * 4, ix->stream, false);
+  if (st == DG_OK)
+    st = dbuf_reserve(d_off, ((size_t)nlist + 2) * 8 + (size_t)nlist * 8,
+                      ix->stream, false);
+  if (st == DG_OK)
+    st = dbuf_reserve(ix->d_centroids, (size_t)nlist * d * 4, ix->stream,
+                      false);
+  if (st == DG_OK)
+    st = dbuf_reserve(ix->d_cnorms, (size_t)nlist * 4, ix->stream, false);
+  if (st != DG_OK) {
+    dbuf_free(d_td);
+    dbuf_free(d_grouped);
+    dbuf_free(d_asg);
+    dbuf_free(d_perm);
+    dbuf_free(d_off);
+    return st;
+  }
+  int64_t* d_offsets = (int64_t*)d_off.p;              // nlist+1
+  int32_t* d_counts = (int32_t*)(d_offsets + nlist + 1);  // nlist (reuse tail)
+  do {
+    if (hipMemcpyAsync(d_td.p, src, (size_t)nt * d * 4, hipMemcpyHostToDevice,
+                       ix->stream) != hipSuccess) {
+      st = DG_EINTERNAL;
+      break;
+    }
+    if (ix->desc.metric == DG_METRIC_COSINE)
+      dgk::normalize_rows(ix->stream, (float*)d_td.p, nt, d);
+
+    // init centroids: first nlist of perm(seed+1) of the (subsampled) set
+    {
+      std::vector<int64_t> p2;
+      rand_perm(p2, nt, seed + 1);
+      p2.resize(nlist);
+      dg_dbuf d_idx{};
+      st = dbuf_reserve(d_idx, (size_t)nlist * 8, ix->stream, false);
+      if (st != DG_OK) break;
+      (void)hipMemcpyAsync(d_idx.p, p2.data(), (size_t)nlist * 8,
+                           hipMemcpyHostToDevice, ix->stream);
+      dgk::gather_rows_by_index(ix->stream, (const float*)d_td.p,
+                                (const int64_t*)d_idx.p, nlist, d,
+                                (float*)ix->d_centroids.p);
+      (void)hipStreamSynchronize(ix->stream);
+      dbuf_free(d_idx);
+    }
+
+    Mt19937 split_rng(seed + 2);
+    std::vector<int32_t> h_counts(nlist);
+    std::vector<float> h_cents((size_t)nlist * d);
+    for (int32_t iter = 0; iter < niter && st == DG_OK; iter++) {
+      dgk::row_norms(ix->stream, (const float*)ix->d_centroids.p, nlist, d,
+                     (float*)ix->d_cnorms.p);
+      st = assign_rows(ix, (const float*)d_td.p, nt, (int32_t*)d_asg.p);
+      if (st != DG_OK) break;
+      (void)hipMemsetAsync(d_counts, 0, (size_t)nlist * 4, ix->stream);
+      dgk::hist_assign(ix->stream, (const int32_t*)d_asg.p, nt, nlist,
+                       d_counts);
+      dgk::excl_scan_i32_to_i64(ix->stream, d_counts, nlist, d_offsets);
+      dgk::init_cursors(ix->stream, d_offsets, nlist, d_counts);  // reuse
+      dgk::scatter_perm(ix->stream, (const int32_t*)d_asg.p, nt, nullptr,
+                        d_counts, (uint32_t*)d_perm.p);
+      dgk::gather_rows(ix->stream, (const float*)d_td.p,
+                       (const uint32_t*)d_perm.p, nt, d, (float*)d_grouped.p);
+      dgk::cluster_means(ix->stream, (const float*)d_grouped.p, d_offsets,
+                         nlist, d, (float*)ix->d_centroids.p);
+      // empty-cluster split on host (nlist x d download/upload; faiss
+      // split_clusters semantics — oracle.c states the rule)
+      std::vector<int64_t> h_off(nlist + 1);
+      (void)hipMemcpyAsync(h_off.data(), d_offsets, (nlist + 1) * 8,
+                           hipMemcpyDeviceToHost, ix->stream);
+      if (hipStreamSynchronize(ix->stream) != hipSuccess) {
+        st = DG_EINTERNAL;
+        break;
+      }
+      std::vector<int64_t> hist(nlist);
+      bool any_empty = false;
+      for (int32_t l = 0; l < nlist; l++) {
+        hist[l] = h_off[l + 1] - h_off[l];
+        if (hist[l] == 0) any_empty = true;
+      }
+      if (any_empty) {
+        (void)hipMemcpy(h_cents.data(), ix->d_centroids.p,
+                        (size_t)nlist * d * 4, hipMemcpyDeviceToHost);
+        const float EPS = 1.0f / 1024.0f;
+        for (int32_t ci = 0; ci < nlist; ci++) {
+          if (hist[ci] != 0) continue;
+          int32_t cj = 0;
+          for (;; cj = (cj + 1) % nlist) {
+            float p = (hist[cj] - 1.0f) / (float)(nt - nlist);
+            if (split_rng.rand_float() < p) break;
+          }
+          memcpy(&h_cents[(size_t)ci * d], &h_cents[(size_t)cj * d],
+                 (size_t)d * 4);
+          for (int32_t j = 0; j < d; j++) {
+            float s = (j % 2 == 0) ? 1 + EPS : 1 - EPS;
+            h_cents[(size_t)ci * d + j] *= s;
+            h_cents[(size_t)cj * d + j] *= 2 - s;
+          }
+          hist[ci] = hist[cj] / 2;
+          hist[cj] -= hist[ci];
+        }
+        (void)hipMemcpy(ix->d_centroids.p, h_cents.data(),
+                        (size_t)nlist * d * 4, hipMemcpyHostToDevice);
+      }
+    }
+    if (st != DG_OK) break;
+    dgk::row_norms(ix->stream, (const float*)ix->d_centroids.p, nlist, d,
+                   (float*)ix->d_cnorms.p);
+    if (hipStreamSynchronize(ix->stream) != hipSuccess) st = DG_EINTERNAL;
+  } while (0);
+  dbuf_free(d_td);
+  dbuf_free(d_grouped);
+  dbuf_free(d_asg);
+  dbuf_free(d_perm);
+  dbuf_free(d_off);
+  if (st == DG_OK) {
+    ix->trained = true;
+    ix->csr_valid = false;
+  }
+  return st;
+}
+
+// ---------------- add / remove ----------------
+extern "C" dg_status dg_add(dg_index* ix, int64_t n, const int64_t* ids,
+                            const float* x) {
+  if (!ix || !ids || !x || n <= 0) {
+    dg_set_error("bad add args");
+    return DG_EINVAL;
+  }
+  std::unique_lock lk(ix->rw);
+  if (!ix->trained) {
+    dg_set_error("IVF index not trained (EVECTOR_NOT_TRAIN)");
+    return DG_ENOT_TRAINED;
+  }
+  // duplicate checks (CheckVectorIdDuplicated + existing-id upsert rule)
+  for (int64_t i = 0; i < n; i++) {
+    if (ids[i] < 0) {
+      dg_set_error("negative id %lld", (long long)ids[i]);
+      return DG_EINVAL;
+    }
+    if (ix->id_count.count(ids[i])) {
+      dg_set_error("vector id duplicated: %lld (EVECTOR_ID_DUPLICATED)",
+                   (long long)ids[i]);
+      return DG_EID_DUPLICATED;
+    }
+  }
+  {
+    std::unordered_map<int64_t, int32_t> batch;
+    for (int64_t i = 0; i < n; i++)
+      if (++batch[ids[i]] > 1) {
+        dg_set_error("vector id duplicated in batch: %lld",
+                     (long long)ids[i]);
+        return DG_EID_DUPLICATED;
+      }
+  }
+  DeviceGuard g(ix->device);
+  const int32_t d = ix->desc.d;
+  const int64_t n0 = ix->ntotal;
+  dg_status st =
+      dbuf_reserve(ix->d_vectors, (size_t)(n0 + n) * d * 4, ix->stream, true);
+  if (st == DG_OK)
+    st = dbuf_reserve(ix->d_ids, (size_t)(n0 + n) * 8, ix->stream, true);
+  if (st == DG_OK)
+    st = dbuf_reserve(ix->d_assign, (size_t)(n0 + n) * 4, ix->stream, true);
+  if (st != DG_OK) return st;
+  float* dst = (float*)ix->d_vectors.p + (size_t)n0 * d;
+  DG_HIP_CHECK(hipMemcpyAsync(dst, x, (size_t)n * d * 4,
+                              hipMemcpyHostToDevice, ix->stream));
+  DG_HIP_CHECK(hipMemcpyAsync((int64_t*)ix->d_ids.p + n0, ids, (size_t)n * 8,
+                              hipMemcpyHostToDevice, ix->stream));
+  if (ix->desc.metric == DG_METRIC_COSINE)
+    dgk::normalize_rows(ix->stream, dst, n, d);
+  if (ix->desc.kind == DG_INDEX_IVF_FLAT) {
+    st = assign_rows(ix, dst, n, (int32_t*)ix->d_assign.p + n0);
+    if (st != DG_OK) return st;
+  } else {
+    dgk::iota_i32(ix->stream, (int32_t*)ix->d_assign.p + n0, n, 0);
+  }
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  for (int64_t i = 0; i < n; i++) ix->id_count.emplace(ids[i], 1);
+  ix->ntotal += n;
+  ix->csr_valid = false;
+  return DG_OK;
+}
+
+extern "C" dg_status dg_remove(dg_index* ix, int64_t n, const int64_t* ids) {
+  if (!ix || !ids || n <= 0) return DG_EINVAL;
+  std::unique_lock lk(ix->rw);
+  // all must exist, else nothing removed (ivf_flat.cc:177-186)
+  for (int64_t i = 0; i < n; i++)
+    if (!ix->id_count.count(ids[i])) {
+      dg_set_error("remove not found vector id %lld (EVECTOR_INVALID)",
+                   (long long)ids[i]);
+      return DG_ENOT_FOUND;
+    }
+  DeviceGuard g(ix->device);
+  // find arrival positions by scanning host mirror of ids?  We keep only a
+  // presence map; positions are found on device: download ids once.
+  std::vector<int64_t> h_ids(ix->ntotal);
+  DG_HIP_CHECK(hipMemcpy(h_ids.data(), ix->d_ids.p, (size_t)ix->ntotal * 8,
+                         hipMemcpyDeviceToHost));
+  std::unordered_map<int64_t, int64_t> pos;
+  pos.reserve(n * 2);
+  for (int64_t i = 0; i < ix->ntotal; i++) pos.emplace(h_ids[i], i);
+  std::vector<int64_t> plist(n);
+  for (int64_t i = 0; i < n; i++) plist[i] = pos[ids[i]];
+  dg_dbuf d_pos{};
+  dg_status st = dbuf_reserve(d_pos, (size_t)n * 8, ix->stream, false);
+  if (st != DG_OK) return st;
+  (void)hipMemcpyAsync(d_pos.p, plist.data(), (size_t)n * 8,
+                       hipMemcpyHostToDevice, ix->stream);
+  dgk::tombstone(ix->stream, (const int64_t*)d_pos.p, n,
+                 (int64_t*)ix->d_ids.p);
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  dbuf_free(d_pos);
+  for (int64_t i = 0; i < n; i++) ix->id_count.erase(ids[i]);
+  ix->n_deleted += n;
+  ix->csr_valid = false;
+  return DG_OK;
+}
+
+extern "C" dg_status dg_upsert(dg_index* ix, int64_t n, const int64_t* ids,
+                               const float* x) {
+  if (!ix || !ids || !x || n <= 0) return DG_EINVAL;
+  std::vector<int64_t> existing;
+  {
+    std::shared_lock lk(ix->rw);
+    for (int64_t i = 0; i < n; i++)
+      if (ix->id_count.count(ids[i])) existing.push_back(ids[i]);
+  }
+  if (!existing.empty()) {
+    dg_status st = dg_remove(ix, existing.size(), existing.data());
+    if (st != DG_OK) return st;
+  }
+  return dg_add(ix, n, ids, x);
+}
+
+extern "C" dg_status dg_set_list_mask(dg_index* ix, const uint8_t* mask) {
+  if (!ix) return DG_EINVAL;
+  std::unique_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  if (!mask) {
+    ix->has_mask = false;
+    return DG_OK;
+  }
+  if (ix->desc.kind != DG_INDEX_IVF_FLAT) {
+    dg_set_error("list mask on non-IVF index");
+    return DG_EINVAL;
+  }
+  dg_status st = dbuf_reserve(ix->d_list_mask, (size_t)ix->desc.nlist,
+                              ix->stream, false);
+  if (st != DG_OK) return st;
+  DG_HIP_CHECK(hipMemcpyAsync(ix->d_list_mask.p, mask, ix->desc.nlist,
+                              hipMemcpyHostToDevice, ix->stream));
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  ix->has_mask = true;
+  return DG_OK;
+}
+
+// ---------------- finalize: arrival arrays -> CSR ----------------
+static dg_status finalize_csr(dg_index* ix) {
+  const int32_t nlist =
+      ix->desc.kind == DG_INDEX_FLAT ? 1 : ix->desc.nlist;
+  const int32_t d = ix->desc.d;
+  const int64_t n = ix->ntotal;
+  dg_status st;
+  if ((st = dbuf_reserve(ix->d_csr_offsets, ((size_t)nlist + 1) * 8,
+                         ix->stream, false)) != DG_OK ||
+      (st = dbuf_reserve(ix->d_csr_vectors, (size_t)n * d * 4, ix->stream,
+                         false)) != DG_OK ||
+      (st = dbuf_reserve(ix->d_csr_ids, (size_t)n * 8, ix->stream, false)) !=
+          DG_OK ||
+      (st = dbuf_reserve(ix->d_csr_vnorms, (size_t)n * 4, ix->stream,
+                         false)) != DG_OK ||
+      (st = dbuf_reserve(ix->ws_small,
+                         (size_t)nlist * 4 + (size_t)n * 4 + 64, ix->stream,
+                         false)) != DG_OK)
+    return st;
+  int32_t* d_counts = (int32_t*)ix->ws_small.p;
+  uint32_t* d_perm = (uint32_t*)((char*)ix->ws_small.p + (size_t)nlist * 4);
+  int64_t* d_offsets = (int64_t*)ix->d_csr_offsets.p;
+  if (n > 0) {
+    (void)hipMemsetAsync(d_counts, 0, (size_t)nlist * 4, ix->stream);
+    dgk::hist_assign(ix->stream, (const int32_t*)ix->d_assign.p, n, nlist,
+                     d_counts);
+    dgk::excl_scan_i32_to_i64(ix->stream, d_counts, nlist, d_offsets);
+    dgk::init_cursors(ix->stream, d_offsets, nlist, d_counts);
+    dgk::scatter_perm(ix->stream, (const int32_t*)ix->d_assign.p, n, nullptr,
+                      d_counts, d_perm);
+    dgk::gather_rows(ix->stream, (const float*)ix->d_vectors.p, d_perm, n, d,
+                     (float*)ix->d_csr_vectors.p);
+    dgk::gather_ids(ix->stream, (const int64_t*)ix->d_ids.p, d_perm, n,
+                    (int64_t*)ix->d_csr_ids.p);
+    dgk::row_norms(ix->stream, (const float*)ix->d_csr_vectors.p, n, d,
+                   (float*)ix->d_csr_vnorms.p);
+  } else {
+    (void)hipMemsetAsync(d_offsets, 0, ((size_t)nlist + 1) * 8, ix->stream);
+  }
+  ix->h_csr_offsets.resize(nlist + 1);
+  DG_HIP_CHECK(hipMemcpyAsync(ix->h_csr_offsets.data(), d_offsets,
+                              ((size_t)nlist + 1) * 8, hipMemcpyDeviceToHost,
+                              ix->stream));
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  ix->csr_valid = true;
+  return DG_OK;
+}
+
+// ---------------- search core ----------------
+static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
+                             int32_t k, int32_t nprobe,
+                             const dg_filter* filter, float* d_out_dist,
+                             int64_t* d_out_ids) {
+  const int32_t d = ix->desc.d;
+  const int metric = ix->desc.metric;
+  const bool is_ivf = ix->desc.kind == DG_INDEX_IVF_FLAT;
+  const int32_t nlist = is_ivf ? ix->desc.nlist : 1;
+  dg_status st = DG_OK;
+
+  (void)hipEventRecord(ix->ev[0], ix->stream);
+
+  // --- queries: copy (never mutate caller buffer), cosine-normalize, norms
+  if ((st = dbuf_reserve(ix->ws_queries, (size_t)nq * d * 4, ix->stream,
+                         false)) != DG_OK ||
+      (st = dbuf_reserve(ix->ws_qnorms, (size_t)nq * 4, ix->stream, false)) !=
+          DG_OK)
+    return st;
+  float* dq = (float*)ix->ws_queries.p;
+  DG_HIP_CHECK(hipMemcpyAsync(dq, d_x, (size_t)nq * d * 4,
+                              hipMemcpyDeviceToDevice, ix->stream));
+  if (metric == DG_METRIC_COSINE) dgk::normalize_rows(ix->stream, dq, nq, d);
+  float* dqn = (float*)ix->ws_qnorms.p;
+  if (metric == DG_METRIC_L2) dgk::row_norms(ix->stream, dq, nq, d, dqn);
+
+  // --- filter -> device structures + row pass bitmap
+  dg_dev_filter df{};
+  df.kind = DG_FILTER_NONE;
+  dg_dbuf d_fids{};
+  bool need_bitmap = ix->n_deleted > 0;
+  if (filter && filter->kind != DG_FILTER_NONE) {
+    df.kind = filter->kind;
+    df.negate = filter->negate;
+    df.min_id = filter->min_id;
+    df.max_id = filter->max_id;
+    need_bitmap = true;
+    if (filter->kind == DG_FILTER_SORTED_IDS) {
+      if ((st = dbuf_reserve(d_fids, (size_t)filter->n_ids * 8, ix->stream,
+                             false)) != DG_OK)
+        return st;
+      (void)hipMemcpyAsync(d_fids.p, filter->ids, (size_t)filter->n_ids * 8,
+                           hipMemcpyHostToDevice, ix->stream);
+      df.ids = (const int64_t*)d_fids.p;
+      df.n_ids = filter->n_ids;
+    } else if (filter->kind == DG_FILTER_BITMAP) {
+      size_t words = (size_t)((filter->bitmap_nbits + 63) / 64);
+      if ((st = dbuf_reserve(d_fids, words * 8, ix->stream, false)) != DG_OK)
+        return st;
+      (void)hipMemcpyAsync(d_fids.p, filter->bitmap, words * 8,
+                           hipMemcpyHostToDevice, ix->stream);
+      df.bitmap = (const uint64_t*)d_fids.p;
+      df.bitmap_base = filter->bitmap_base;
+      df.bitmap_nbits = filter->bitmap_nbits;
+    }
+  }
+  uint32_t* d_bitmap = nullptr;
+  dg_dbuf ws_bitmap{};
+  if (need_bitmap && ix->ntotal > 0) {
+    size_t words = (size_t)((ix->ntotal + 31) / 32);
+    if ((st = dbuf_reserve(ws_bitmap, words * 4, ix->stream, false)) != DG_OK) {
+      dbuf_free(d_fids);
+      return st;
+    }
+    dgk::build_pass_bitmap(ix->stream, (const int64_t*)ix->d_csr_ids.p,
+                           ix->ntotal, &df, (uint32_t*)ws_bitmap.p);
+    d_bitmap = (uint32_t*)ws_bitmap.p;
+  }
+
+  if (!is_ivf) {
+    // ---------- FLAT: chunked dots GEMM + dense select ----------
+    const int64_t N = ix->ntotal;
+    const int64_t max_dots = (int64_t)(2048ull << 20) / 4;  // 2 GiB of f32
+    int64_t chunk_cols =
+        std::max<int64_t>(65536, std::min<int64_t>(N, max_dots / nq));
+    int64_t nchunks = (N + chunk_cols - 1) / chunk_cols;
+    if ((st = dbuf_reserve(ix->ws_dots,
+                           (size_t)nq * std::min(N, chunk_cols) * 4,
+                           ix->stream, false)) != DG_OK ||
+        (st = dbuf_reserve(ix->ws_topk,
+                           (size_t)nq * nchunks * k * 8 + (size_t)nq * k * 8 +
+                               (size_t)(nq + 1) * 16,
+                           ix->stream, false)) != DG_OK) {
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      return st;
+    }
+    uint64_t* slab = (uint64_t*)ix->ws_topk.p;          // nq x nchunks*k
+    uint64_t* final_tk = slab + (size_t)nq * nchunks * k;  // nq x k
+    int64_t* bt = (int64_t*)(final_tk + (size_t)nq * k);   // base/total
+    (void)hipEventRecord(ix->ev[1], ix->stream);  // no coarse stage in Flat
+    (void)hipEventRecord(ix->ev[2], ix->stream);
+    for (int64_t ci = 0; ci < nchunks; ci++) {
+      int64_t c0 = ci * chunk_cols;
+      int64_t cc = std::min(chunk_cols, N - c0);
+      st = sgemm_dots(ix, dq, nq,
+                      (const float*)ix->d_csr_vectors.p + (size_t)c0 * d, cc,
+                      d, (float*)ix->ws_dots.p);
+      if (st != DG_OK) break;
+      dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
+                        (const float*)ix->d_csr_vnorms.p + c0, nq, cc, k,
+                        metric == DG_METRIC_L2 ? 1 : 2, d_bitmap, c0, slab,
+                        nchunks * k, ci * k);
+    }
+    (void)hipEventRecord(ix->ev[3], ix->stream);
+    if (st == DG_OK) {
+      uint64_t* result = slab;
+      if (nchunks > 1) {
+        dgk::fill_base_total(ix->stream, nq, nchunks * k, bt, bt + nq);
+        dgk::select_u64(ix->stream, slab, bt, bt + nq, nq, k, final_tk, k);
+        result = final_tk;
+      }
+      dgk::emit_results(ix->stream, result, (const int64_t*)ix->d_csr_ids.p,
+                        dqn, nq, k, metric, metric == DG_METRIC_L2 ? 1 : 0,
+                        d_out_dist, d_out_ids);
+    }
+    (void)hipEventRecord(ix->ev[4], ix->stream);
+    ix->times.last_scan_bytes_alg = (int64_t)ix->ntotal * (d * 4 + 4);
+  } else {
+    // ---------- IVF ----------
+    // nprobe default + clamp (ivf_flat.cc:208-214, :234; default 80 =
+    // kSearchIvfFlatParamNprobe)
+    int32_t np = nprobe > 0 ? nprobe : 80;
+    np = std::min(np, nlist);
+    // coarse: dots + select top-np + unpack (mask applied)
+    if ((st = dbuf_reserve(ix->ws_dots, (size_t)nq * nlist * 4, ix->stream,
+                           false)) != DG_OK ||
+        (st = dbuf_reserve(ix->ws_probes,
+                           (size_t)nq * np * 12 + (size_t)nq * k * 8,
+                           ix->stream, false)) != DG_OK) {
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      return st;
+    }
+    uint64_t* coarse_tk = (uint64_t*)ix->ws_probes.p;        // nq x np
+    int32_t* probes = (int32_t*)(coarse_tk + (size_t)nq * np);  // nq x np
+    uint64_t* final_tk = (uint64_t*)(probes + (size_t)nq * np);  // nq x k
+    st = sgemm_dots(ix, dq, nq, (const float*)ix->d_centroids.p, nlist, d,
+                    (float*)ix->ws_dots.p);
+    if (st != DG_OK) {
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      return st;
+    }
+    dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
+                      (const float*)ix->d_cnorms.p, nq, nlist, np,
+                      metric == DG_METRIC_L2 ? 1 : 2, nullptr, 0, coarse_tk,
+                      np, 0);
+    dgk::probe_unpack(ix->stream, coarse_tk, nq, np,
+                      ix->has_mask ? (const uint8_t*)ix->d_list_mask.p
+                                   : nullptr,
+                      probes);
+    (void)hipEventRecord(ix->ev[1], ix->stream);
+
+    // inverted mapping + candidate offsets
+    size_t inv_bytes = (size_t)nlist * 4 * 3 + ((size_t)nlist + 1) * 12 +
+                       (size_t)nq * np * 8 * 2 + ((size_t)nq + 1) * 8 * 2 +
+                       (size_t)nq * np * 8 + 64;
+    if ((st = dbuf_reserve(ix->ws_inv, inv_bytes, ix->stream, false)) !=
+        DG_OK) {
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      return st;
+    }
+    char* wp = (char*)ix->ws_inv.p;
+    int32_t* inv_counts = (int32_t*)wp;            wp += (size_t)nlist * 4;
+    int32_t* cursors = (int32_t*)wp;               wp += (size_t)nlist * 4;
+    int32_t* unit_counts = (int32_t*)wp;           wp += (size_t)nlist * 4;
+    int64_t* inv_offsets64 = (int64_t*)wp;         wp += ((size_t)nlist + 1) * 8;
+    int32_t* inv_offsets32 = (int32_t*)wp;         wp += ((size_t)nlist + 1) * 4;
+    int32_t* inv_q = (int32_t*)wp;                 wp += (size_t)nq * np * 4;
+    int32_t* inv_rank = (int32_t*)wp;              wp += (size_t)nq * np * 4;
+    int64_t* qp_off = (int64_t*)wp;                wp += (size_t)nq * np * 8;
+    int64_t* q_total = (int64_t*)wp;               wp += (size_t)nq * 8;
+    int64_t* q_cand_base = (int64_t*)wp;  // nq+1
+
+    (void)hipMemsetAsync(inv_counts, 0, (size_t)nlist * 4, ix->stream);
+    dgk::hist_probes(ix->stream, probes, nq, np, nlist, inv_counts);
+    dgk::excl_scan_i32_to_i64(ix->stream, inv_counts, nlist, inv_offsets64);
+    dgk::init_cursors(ix->stream, inv_offsets64, nlist + 1, inv_offsets32);
+    dgk::init_cursors(ix->stream, inv_offsets64, nlist, cursors);
+    dgk::scatter_probes(ix->stream, probes, nq, np, nullptr, cursors, inv_q,
+                        inv_rank);
+    dgk::cand_offsets(ix->stream, probes, nq, np,
+                      (const int64_t*)ix->d_csr_offsets.p, qp_off, q_total);
+    dgk::excl_scan_i64(ix->stream, q_total, nq, q_cand_base);
+    // units
+    const int32_t chunk_rows = 1024;
+    dgk::fill_unit_counts(ix->stream, inv_counts, nlist,
+                          (const int64_t*)ix->d_csr_offsets.p, chunk_rows,
+                          unit_counts);
+    // reuse cursors space scan: unit offsets (i64 in ws, reuse inv_offsets?
+    // inv_offsets64 still needed by scan kernel; allocate unit offsets in
+    // ws_units along with the unit array)
+    // sizing sync: total candidates + total units + inv_counts (for stats)
+    dg_dbuf& wsu = ix->ws_units;
+    if ((st = dbuf_reserve(wsu, ((size_t)nlist + 1) * 8, ix->stream, false))
+        != DG_OK) {
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      return st;
+    }
+    int64_t* unit_offsets = (int64_t*)wsu.p;
+    dgk::excl_scan_i32_to_i64(ix->stream, unit_counts, nlist, unit_offsets);
+    int64_t totals[2] = {0, 0};
+    std::vector<int32_t> h_inv_counts(nlist);
+    DG_HIP_CHECK(hipMemcpyAsync(&totals[0], q_cand_base + nq, 8,
+                                hipMemcpyDeviceToHost, ix->stream));
+    DG_HIP_CHECK(hipMemcpyAsync(&totals[1], unit_offsets + nlist, 8,
+                                hipMemcpyDeviceToHost, ix->stream));
+    DG_HIP_CHECK(hipMemcpyAsync(h_inv_counts.data(), inv_counts,
+                                (size_t)nlist * 4, hipMemcpyDeviceToHost,
+                                ix->stream));
+    DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+    int64_t total_cand = totals[0];
+    int32_t total_units = (int32_t)totals[1];
+    // algorithmic bytes: one read of each probed list (vectors + norms)
+    int64_t alg = 0;
+    for (int32_t l = 0; l < nlist; l++)
+      if (h_inv_counts[l] > 0)
+        alg += (ix->h_csr_offsets[l + 1] - ix->h_csr_offsets[l]) *
+               (int64_t)(d * 4 + 4);
+    ix->times.last_scan_bytes_alg = alg;
+
+    size_t need = ((size_t)nlist + 1) * 8 + (size_t)total_units * 8;
+    if ((st = dbuf_reserve(wsu, need, ix->stream, true)) != DG_OK ||
+        (st = dbuf_reserve(ix->ws_cand, (size_t)total_cand * 8, ix->stream,
+                           false)) != DG_OK) {
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      return st;
+    }
+    unit_offsets = (int64_t*)wsu.p;
+    uint32_t* units = (uint32_t*)(unit_offsets + nlist + 1);
+    // fill units from unit_offsets (i64 -> i32 via cursors copy)
+    dgk::init_cursors(ix->stream, unit_offsets, nlist, cursors);
+    dgk::fill_units(ix->stream, cursors, unit_counts, nlist, nullptr, 0,
+                    units, total_units);
+
+    // THE scan
+    int32_t qt_max = std::max(1, std::min(16, (int32_t)(131072 / (4 * d))));
+    (void)hipEventRecord(ix->ev[2], ix->stream);
+    dgk::ivf_scan(ix->stream, units, total_units,
+                  (const int64_t*)ix->d_csr_offsets.p,
+                  (const float*)ix->d_csr_vectors.p,
+                  (const float*)ix->d_csr_vnorms.p, nullptr, dq, nullptr, d,
+                  inv_offsets32, inv_q, inv_rank, qp_off, q_cand_base, metric,
+                  d_bitmap, chunk_rows, np, qt_max,
+                  (uint64_t*)ix->ws_cand.p);
+    (void)hipEventRecord(ix->ev[3], ix->stream);
+
+    // select + emit
+    dgk::select_u64(ix->stream, (const uint64_t*)ix->ws_cand.p, q_cand_base,
+                    q_total, nq, k, final_tk, k);
+    dgk::emit_results(ix->stream, final_tk, (const int64_t*)ix->d_csr_ids.p,
+                      dqn, nq, k, metric, metric == DG_METRIC_L2 ? 1 : 0,
+                      d_out_dist, d_out_ids);
+    (void)hipEventRecord(ix->ev[4], ix->stream);
+  }
+  dbuf_free(d_fids);
+  dbuf_free(ws_bitmap);
+  ix->times.last_nq = nq;
+  return st;
+}
+
+// NOTE: ivf_scan needs inv_offsets (int32) — computed as `cursors` would be
+// overwritten; see fix in search_core (we pass a dedicated buffer).
+
+extern "C" dg_status dg_search_device(dg_index* ix, int64_t nq,
+                                      const float* d_x, int32_t k,
+                                      int32_t nprobe, const dg_filter* filter,
+                                      float* d_out_dist, int64_t* d_out_ids) {
+  if (!ix || !d_x || !d_out_dist || !d_out_ids || nq <= 0) {
+    dg_set_error("bad search args");
+    return DG_EINVAL;
+  }
+  if (k <= 0) return DG_OK;  // reference: topk <= 0 => OK no-op
+  if (k > 128) {
+    dg_set_error("k > 128 not supported this round");
+    return DG_ENOT_SUPPORT;
+  }
+  DeviceGuard g(ix->device);
+  // untrained IVF: blank results, OK (ivf_flat.cc:223-227)
+  if (!ix->trained || ix->ntotal == 0) {
+    std::shared_lock lk(ix->rw);
+    (void)hipMemsetAsync(d_out_dist, 0, (size_t)nq * k * 4, ix->stream);
+    (void)hipMemsetAsync(d_out_ids, 0xff, (size_t)nq * k * 8, ix->stream);
+    return DG_OK;
+  }
+  {
+    std::unique_lock lk(ix->rw, std::defer_lock);
+    if (!ix->csr_valid) {
+      lk.lock();
+      if (!ix->csr_valid) {
+        dg_status st = finalize_csr(ix);
+        if (st != DG_OK) return st;
+      }
+    }
+  }
+  std::shared_lock lk(ix->rw);
+  return search_core(ix, nq, d_x, k, nprobe, filter, d_out_dist, d_out_ids);
+}
+
+extern "C" dg_status dg_search(dg_index* ix, int64_t nq, const float* x,
+                               int32_t k, int32_t nprobe,
+                               const dg_filter* filter, float* out_dist,
+                               int64_t* out_ids) {
+  if (!ix || !x || !out_dist || !out_ids || nq <= 0) {
+    dg_set_error("bad search args");
+    return DG_EINVAL;
+  }
+  if (k <= 0) return DG_OK;
+  DeviceGuard g(ix->device);
+  // upload queries, run device path, download
+  static thread_local dg_dbuf t_in, t_dist, t_ids;  // per-thread staging
+  dg_status st;
+  if ((st = dbuf_reserve(t_in, (size_t)nq * ix->desc.d * 4, ix->stream,
+                         false)) != DG_OK ||
+      (st = dbuf_reserve(t_dist, (size_t)nq * k * 4, ix->stream, false)) !=
+          DG_OK ||
+      (st = dbuf_reserve(t_ids, (size_t)nq * k * 8, ix->stream, false)) !=
+          DG_OK)
+    return st;
+  DG_HIP_CHECK(hipMemcpyAsync(t_in.p, x, (size_t)nq * ix->desc.d * 4,
+                              hipMemcpyHostToDevice, ix->stream));
+  st = dg_search_device(ix, nq, (const float*)t_in.p, k, nprobe, filter,
+                        (float*)t_dist.p, (int64_t*)t_ids.p);
+  if (st != DG_OK) return st;
+  DG_HIP_CHECK(hipMemcpyAsync(out_dist, t_dist.p, (size_t)nq * k * 4,
+                              hipMemcpyDeviceToHost, ix->stream));
+  DG_HIP_CHECK(hipMemcpyAsync(out_ids, t_ids.p, (size_t)nq * k * 8,
+                              hipMemcpyDeviceToHost, ix->stream));
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  return DG_OK;
+}
+
+extern "C" dg_status dg_sync(dg_index* ix) {
+  if (!ix) return DG_EINVAL;
+  DeviceGuard g(ix->device);
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  return DG_OK;
+}
+
+// ---------------- range search: §8f rank 2, not yet ----------------
+extern "C" dg_status dg_range_search(dg_index*, int64_t, const float*, float,
+                                     const dg_filter*, int64_t*, int64_t**,
+                                     float**) {
+  dg_set_error("range search lands with SURVEY.md §8f rank 2 "
+               "(EVECTOR_NOT_SUPPORT => reader brute-force fallback)");
+  return DG_ENOT_SUPPORT;
+}
+
+extern "C" void dg_free(void* p) { free(p); }
+
+// ---------------- save / load (container v1) ----------------
+static const uint32_t kMagic = 0x44474931;  // "DGI1"
+
+extern "C" dg_status dg_save(dg_index* ix, const char* path) {
+  if (!ix || !path) return DG_EINVAL;
+  std::shared_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  FILE* f = fopen(path, "wb");
+  if (!f) {
+    dg_set_error("cannot open %s", path);
+    return DG_EIO;
+  }
+  const int32_t d = ix->desc.d;
+  int32_t trained = ix->trained ? 1 : 0;
+  fwrite(&kMagic, 4, 1, f);
+  fwrite(&ix->desc, sizeof(ix->desc), 1, f);
+  fwrite(&trained, 4, 1, f);
+  fwrite(&ix->ntotal, 8, 1, f);
+  fwrite(&ix->n_deleted, 8, 1, f);
+  dg_status st = DG_OK;
+  if (ix->desc.kind == DG_INDEX_IVF_FLAT && ix->trained) {
+    std::vector<float> cents((size_t)ix->desc.nlist * d);
+    if (hipMemcpy(cents.data(), ix->d_centroids.p, cents.size() * 4,
+                  hipMemcpyDeviceToHost) != hipSuccess)
+      st = DG_EINTERNAL;
+    fwrite(cents.data(), 4, cents.size(), f);
+  }
+  if (st == DG_OK && ix->ntotal > 0) {
+    const size_t CH = 1 << 20;  // rows per host staging chunk
+    std::vector<float> vbuf(CH * d);
+    std::vector<int64_t> ibuf(CH);
+    std::vector<int32_t> abuf(CH);
+    for (int64_t s0 = 0; s0 < ix->ntotal && st == DG_OK; s0 += CH) {
+      size_t c = std::min<int64_t>(CH, ix->ntotal - s0);
+      if (hipMemcpy(vbuf.data(), (float*)ix->d_vectors.p + (size_t)s0 * d,
+                    c * d * 4, hipMemcpyDeviceToHost) != hipSuccess ||
+          hipMemcpy(ibuf.data(), (int64_t*)ix->d_ids.p + s0, c * 8,
+                    hipMemcpyDeviceToHost) != hipSuccess ||
+          hipMemcpy(abuf.data(), (int32_t*)ix->d_assign.p + s0, c * 4,
+                    hipMemcpyDeviceToHost) != hipSuccess)
+        st = DG_EINTERNAL;
+      fwrite(vbuf.data(), 4, c * d, f);
+      fwrite(ibuf.data(), 8, c, f);
+      fwrite(abuf.data(), 4, c, f);
+    }
+  }
+  fclose(f);
+  return st;
+}
+
+extern "C" dg_status dg_load(dg_index** out, const char* path,
+                             int32_t device) {
+  if (!out || !path) return DG_EINVAL;
+  FILE* f = fopen(path, "rb");
+  if (!f) {
+    dg_set_error("cannot open %s", path);
+    return DG_EIO;
+  }
+  uint32_t magic = 0;
+  dg_index_desc desc{};
+  int32_t trained = 0;
+  int64_t ntotal = 0, ndel = 0;
+  if (fread(&magic, 4, 1, f) != 1 || magic != kMagic ||
+      fread(&desc, sizeof(desc), 1, f) != 1 || fread(&trained, 4, 1, f) != 1 ||
+      fread(&ntotal, 8, 1, f) != 1 || fread(&ndel, 8, 1, f) != 1) {
+    fclose(f);
+    dg_set_error("bad container header in %s", path);
+    return DG_EIO;
+  }
+  desc.device = device;
+  dg_index* ix = nullptr;
+  dg_status st = dg_index_create(&ix, &desc);
+  if (st != DG_OK) {
+    fclose(f);
+    return st;
+  }
+  DeviceGuard g(ix->device);
+  const int32_t d = desc.d;
+  do {
+    if (desc.kind == DG_INDEX_IVF_FLAT && trained) {
+      std::vector<float> cents((size_t)desc.nlist * d);
+      if (fread(cents.data(), 4, cents.size(), f) != cents.size()) {
+        st = DG_EIO;
+        break;
+      }
+      st = dg_set_centroids(ix, desc.nlist, cents.data());
+      if (st != DG_OK) break;
+    }
+    if (ntotal > 0) {
+      if ((st = dbuf_reserve(ix->d_vectors, (size_t)ntotal * d * 4,
+                             ix->stream, false)) != DG_OK ||
+          (st = dbuf_reserve(ix->d_ids, (size_t)ntotal * 8, ix->stream,
+                             false)) != DG_OK ||
+          (st = dbuf_reserve(ix->d_assign, (size_t)ntotal * 4, ix->stream,
+                             false)) != DG_OK)
+        break;
+      const size_t CH = 1 << 20;
+      std::vector<float> vbuf(CH * d);
+      std::vector<int64_t> ibuf(CH);
+      std::vector<int32_t> abuf(CH);
+      for (int64_t s0 = 0; s0 < ntotal && st == DG_OK; s0 += CH) {
+        size_t c = std::min<int64_t>(CH, ntotal - s0);
+        if (fread(vbuf.data(), 4, c * d, f) != c * d ||
+            fread(ibuf.data(), 8, c, f) != c ||
+            fread(abuf.data(), 4, c, f) != c) {
+          st = DG_EIO;
+          break;
+        }
+        if (hipMemcpy((float*)ix->d_vectors.p + (size_t)s0 * d, vbuf.data(),
+                      c * d * 4, hipMemcpyHostToDevice) != hipSuccess ||
+            hipMemcpy((int64_t*)ix->d_ids.p + s0, ibuf.data(), c * 8,
+                      hipMemcpyHostToDevice) != hipSuccess ||
+            hipMemcpy((int32_t*)ix->d_assign.p + s0, abuf.data(), c * 4,
+                      hipMemcpyHostToDevice) != hipSuccess)
+          st = DG_EINTERNAL;
+        for (size_t i = 0; i < c && st == DG_OK; i++)
+          if (ibuf[i] >= 0) ix->id_count.emplace(ibuf[i], 1);
+      }
+      ix->ntotal = ntotal;
+      ix->n_deleted = ndel;
+      ix->csr_valid = false;
+    }
+  } while (0);
+  fclose(f);
+  if (st != DG_OK) {
+    dg_index_destroy(ix);
+    return st;
+  }
+  ix->trained = trained != 0;
+  *out = ix;
+  return DG_OK;
+}
+
+// ---------------- stats ----------------
+extern "C" dg_status dg_stats(dg_index* ix, dg_stats_out* out) {
+  if (!ix || !out) return DG_EINVAL;
+  std::shared_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  memset(out, 0, sizeof(*out));
+  out->ntotal = ix->ntotal - ix->n_deleted;
+  out->d = ix->desc.d;
+  out->metric = ix->desc.metric;
+  out->kind = ix->desc.kind;
+  out->nlist = ix->desc.nlist;
+  out->is_trained = ix->trained ? 1 : 0;
+  size_t db = 0;
+  for (auto b : {&ix->d_vectors, &ix->d_ids, &ix->d_assign, &ix->d_centroids,
+                 &ix->d_cnorms, &ix->d_csr_offsets, &ix->d_csr_vectors,
+                 &ix->d_csr_ids, &ix->d_csr_vnorms})
+    db += b->cap;
+  out->device_bytes = (int64_t)db;
+  if (ix->times.last_nq > 0) {
+    (void)hipEventSynchronize(ix->ev[4]);
+    float ms01 = 0, ms23 = 0, ms04 = 0;
+    (void)hipEventElapsedTime(&ms01, ix->ev[0], ix->ev[1]);
+    (void)hipEventElapsedTime(&ms23, ix->ev[2], ix->ev[3]);
+    (void)hipEventElapsedTime(&ms04, ix->ev[0], ix->ev[4]);
+    out->last_coarse_ms = ms01;
+    out->last_scan_ms = ms23;
+    out->last_total_ms = ms04;
+    out->last_select_ms = ms04 - ms23 - ms01;
+    out->last_nq = ix->times.last_nq;
+    out->last_scan_bytes_algorithmic = ix->times.last_scan_bytes_alg;
+    if (ms23 > 0)
+      out->last_scan_gbps_algorithmic =
+          (double)ix->times.last_scan_bytes_alg / (ms23 * 1e6);
+  }
+  return DG_OK;
+}

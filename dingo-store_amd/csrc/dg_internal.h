@@ -1,0 +1,195 @@
+// dg_internal.h — internal state of the MI355X-native index library.
+// Product path: no oracle code is linked or called here (oracle/ is test
+// infrastructure only).  All compute runs on the GPU; there is no CPU
+// fallback — entry points fail with DG_ENOGPU when no device is present.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <rocblas/rocblas.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <mutex>
+#include <shared_mutex>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "../../include/dingo_gpu.h"
+
+#define DG_HIP_CHECK(expr)                                              \
+  do {                                                                  \
+    hipError_t _e = (expr);                                             \
+    if (_e != hipSuccess) {                                             \
+      dg_set_error("HIP error %s at %s:%d: %s", hipGetErrorName(_e),    \
+                   __FILE__, __LINE__, hipGetErrorString(_e));          \
+      return DG_EINTERNAL;                                              \
+    }                                                                   \
+  } while (0)
+
+#define DG_ROCBLAS_CHECK(expr)                                          \
+  do {                                                                  \
+    rocblas_status _s = (expr);                                         \
+    if (_s != rocblas_status_success) {                                 \
+      dg_set_error("rocBLAS error %d at %s:%d", (int)_s, __FILE__,      \
+                   __LINE__);                                           \
+      return DG_EINTERNAL;                                              \
+    }                                                                   \
+  } while (0)
+
+void dg_set_error(const char* fmt, ...);
+
+// Device-side filter descriptor (POD copied into kernel args).
+struct dg_dev_filter {
+  int32_t kind;    // dg_filter_kind
+  int32_t negate;
+  int64_t min_id, max_id;
+  const int64_t* ids;       // device pointer (uploaded), ascending
+  int64_t n_ids;
+  const uint64_t* bitmap;   // device pointer
+  int64_t bitmap_base, bitmap_nbits;
+};
+
+// A growable device buffer.
+struct dg_dbuf {
+  void* p = nullptr;
+  size_t bytes = 0, cap = 0;
+};
+
+struct dg_stage_times {
+  double coarse_ms = 0, scan_ms = 0, select_ms = 0, total_ms = 0;
+  // cumulative since last reset (for bench averaging)
+  double acc_scan_ms = 0;
+  int64_t acc_scan_launches = 0;
+  int64_t last_scan_bytes_alg = 0;
+  int64_t last_nq = 0;
+};
+
+struct dg_index {
+  dg_index_desc desc;
+  int device = 0;
+  hipStream_t stream = nullptr;
+  rocblas_handle blas = nullptr;
+
+  // ---- store (arrival order; the base of truth) ----
+  dg_dbuf d_vectors;   // [ntotal x d] f32 (normalized already for cosine)
+  dg_dbuf d_ids;       // [ntotal] i64
+  dg_dbuf d_assign;    // [ntotal] i32 (IVF only; coarse list per vector)
+  int64_t ntotal = 0;
+  int64_t n_deleted = 0;
+
+  // ---- IVF structure ----
+  bool trained = false;
+  dg_dbuf d_centroids;      // [nlist x d]
+  dg_dbuf d_cnorms;         // [nlist] f32  (L2 coarse via norms trick)
+  // finalized CSR (built lazily from arrival store on first search)
+  bool csr_valid = false;
+  dg_dbuf d_csr_offsets;    // [nlist+1] i64
+  dg_dbuf d_csr_vectors;    // [ntotal x d] grouped by list
+  dg_dbuf d_csr_ids;        // [ntotal] i64 (-2 = deleted tombstone)
+  dg_dbuf d_csr_vnorms;     // [ntotal] f32 norms of grouped vectors
+  std::vector<int64_t> h_csr_offsets;  // host copy for planning
+
+  // Flat uses d_csr_* with a single implicit list (nlist=1) so scan/select
+  // machinery is shared.
+
+  // optional list ownership mask (multi-GPU list sharding)
+  dg_dbuf d_list_mask;      // [nlist] u8, empty = all owned
+  bool has_mask = false;
+
+  // id -> present (host, duplicate detection / remove)
+  std::unordered_map<int64_t, int32_t> id_count;
+
+  // workspaces (grown on demand, reused across searches)
+  dg_dbuf ws_queries, ws_qnorms, ws_dots, ws_probes, ws_inv, ws_cand,
+      ws_units, ws_small, ws_topk;
+
+  // timing
+  hipEvent_t ev[12] = {};
+  dg_stage_times times;
+  bool events_ready = false;
+
+  std::shared_mutex rw;  // search shared; mutation exclusive
+};
+
+// ---- kernel launchers (kernels.hip.cpp; authoritative signatures) ----
+namespace dgk {
+void probe_unpack(hipStream_t s, const uint64_t* topk, int64_t nq,
+                  int32_t nprobe, const uint8_t* mask, int32_t* probes);
+void init_cursors(hipStream_t s, const int64_t* offsets, int32_t n,
+                  int32_t* cursors);
+void select_dense(hipStream_t s, const float* scores, const float* cnorms,
+                  int64_t rows, int64_t cols, int32_t k, int mode,
+                  const uint32_t* bitmap, int64_t col_base, uint64_t* out,
+                  int64_t out_stride, int64_t out_offset);
+void select_u64(hipStream_t s, const uint64_t* cand, const int64_t* base,
+                const int64_t* total, int64_t nq, int32_t k, uint64_t* out,
+                int64_t out_stride);
+void gather_rows_by_index(hipStream_t s, const float* src, const int64_t* idx,
+                          int64_t n, int32_t d, float* dst);
+void fill_base_total(hipStream_t s, int64_t nq, int64_t len, int64_t* base,
+                     int64_t* total);
+void tombstone(hipStream_t s, const int64_t* pos, int64_t n, int64_t* ids);
+void row_norms(hipStream_t s, const float* x, int64_t n, int32_t d,
+               float* out);
+void normalize_rows(hipStream_t s, float* x, int64_t n, int32_t d);
+void argmin_rows(hipStream_t s, const float* dots, const float* cnorms,
+                 int64_t n, int32_t nlist, int metric, int32_t* out);
+// build per-row pass bitmap from filter over ids[row] (also kills tombstones)
+void build_pass_bitmap(hipStream_t s, const int64_t* ids, int64_t n,
+                       const dg_dev_filter* f, uint32_t* bitmap);
+// IVF probe machinery
+void probe_mask_apply(hipStream_t s, int32_t* probes, int64_t nq,
+                      int32_t nprobe, const uint8_t* mask);
+void hist_probes(hipStream_t s, const int32_t* probes, int64_t nq,
+                 int32_t nprobe, int32_t nlist, int32_t* counts);
+void scatter_probes(hipStream_t s, const int32_t* probes, int64_t nq,
+                    int32_t nprobe, const int32_t* inv_offsets,
+                    int32_t* cursors, int32_t* inv_q, int32_t* inv_rank);
+void cand_offsets(hipStream_t s, const int32_t* probes, int64_t nq,
+                  int32_t nprobe, const int64_t* csr_offsets,
+                  int64_t* qp_off /* nq*nprobe, offset within query */,
+                  int64_t* q_total /* nq */);
+void fill_unit_counts(hipStream_t s, const int32_t* inv_counts, int32_t nlist,
+                      const int64_t* csr_offsets, int32_t chunk_rows,
+                      int32_t* unit_counts);
+void fill_units(hipStream_t s, const int32_t* unit_offsets,
+                const int32_t* inv_counts, int32_t nlist,
+                const int64_t* csr_offsets, int32_t chunk_rows,
+                uint32_t* units /* 2 x u32 per unit: list, chunk */,
+                int32_t total_units);
+// THE dominant kernel: grouped inverted-list scan.
+void ivf_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
+              const int64_t* csr_offsets, const float* csr_vectors,
+              const float* csr_vnorms, const int64_t* csr_ids,
+              const float* queries, const float* qnorms, int32_t d,
+              const int32_t* inv_offsets, const int32_t* inv_q,
+              const int32_t* inv_rank, const int64_t* qp_off,
+              const int64_t* q_cand_base, int metric,
+              const uint32_t* pass_bitmap, int32_t chunk_rows, int32_t nprobe,
+              int32_t qt_max, uint64_t* cand);
+// emit: resolve ids, apply metric convention
+void emit_results(hipStream_t s, const uint64_t* topk,
+                  const int64_t* ids_lookup, const float* qnorms, int64_t nq,
+                  int32_t k, int metric, int add_qnorm, float* out_dist,
+                  int64_t* out_ids);
+// k-means / finalize helpers
+void hist_assign(hipStream_t s, const int32_t* assign, int64_t n,
+                 int32_t nlist, int32_t* counts);
+void scatter_perm(hipStream_t s, const int32_t* assign, int64_t n,
+                  const int64_t* offsets_i64, int32_t* cursors32,
+                  uint32_t* perm /* dest slot for row i */);
+void gather_rows(hipStream_t s, const float* src, const uint32_t* perm,
+                 int64_t n, int32_t d, float* dst);
+void gather_ids(hipStream_t s, const int64_t* src, const uint32_t* perm,
+                int64_t n, int64_t* dst);
+void cluster_means(hipStream_t s, const float* grouped, const int64_t* offsets,
+                   int32_t nlist, int32_t d, float* centroids);
+void iota_i32(hipStream_t s, int32_t* p, int64_t n, int32_t value);
+void excl_scan_i32_to_i64(hipStream_t s, const int32_t* in, int32_t n,
+                          int64_t* out /* n+1 */);
+void excl_scan_i32(hipStream_t s, const int32_t* in, int32_t n,
+                   int32_t* out /* n+1 */);
+void excl_scan_i64(hipStream_t s, const int64_t* in, int64_t n,
+                   int64_t* out /* n+1 */);
+}  // namespace dgk

@@ -166,7 +166,8 @@ dg_status dg_remove(dg_index* idx, int64_t n, const int64_t* ids);
  * Host-pointer form: copies in/out and synchronizes. */
 dg_status dg_search(dg_index* idx, int64_t nq, const float* x, int32_t k,
                     int32_t nprobe, const dg_filter* filter,
-                    float* out_dist /* nq*k */, int64_t* out_ids /* nq*k */);
+                    float* out_dist /* nq x k */,
+                    int64_t* out_ids /* nq x k */);
 /* Device-pointer form: x/out_dist/out_ids are device pointers on the index's
  * device; enqueues on the index stream, no synchronization (call dg_sync).
  * This is the resident-in-HBM hot path bench.py times. */

@@ -1,0 +1,376 @@
+"""GPU parity tests (marked gpu): the product HIP path vs the CPU oracle on
+identical seeded inputs, per the parity protocol of SURVEY.md §8c:
+  Flat: identical id sets (fp-tie slack), distances <= 1e-3 abs/rel.
+  IVF : identical centroids injected on both sides, then identical id sets
+        modulo fp-tie reordering; recall vs exhaustive ground truth equal.
+Everything here calls through the C-ABI (ctypes); nothing reads
+/root/reference at run time.
+"""
+import os
+import sys
+
+import numpy as np
+import pytest
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+REPO = os.path.dirname(HERE)
+sys.path[:0] = [REPO, os.path.join(REPO, "oracle"),
+                os.path.join(REPO, "dingo-store_amd")]
+
+import pyoracle as orc  # noqa: E402
+import workload  # noqa: E402
+
+pytestmark = pytest.mark.gpu
+
+dg = pytest.importorskip("dingostore")
+
+rng = np.random.default_rng(20260915)
+
+
+def ids_match_with_tie_slack(gd, gi, od, oi, tol=1e-3):
+    """Rows may disagree only where distances tie within tol."""
+    exact = (gi == oi)
+    if exact.all():
+        return 1.0
+    bad = 0
+    for r in range(gi.shape[0]):
+        if exact[r].all():
+            continue
+        # the multisets of distances must still agree
+        if not np.allclose(np.sort(gd[r]), np.sort(od[r]), rtol=tol,
+                           atol=tol):
+            bad += 1
+            continue
+        # mismatched positions must be fp ties: distance diff within tol
+        for c in np.nonzero(~exact[r])[0]:
+            if abs(gd[r, c] - od[r, c]) > tol * max(1.0, abs(od[r, c])):
+                bad += 1
+                break
+    return 1.0 - bad / gi.shape[0]
+
+
+def make_data(n=20000, d=128, nq=64, seed=777):
+    base = workload.gen_base(seed, n, d)
+    q = workload.gen_queries(seed, n, d, nq)
+    return base, q
+
+
+# ---------------- Flat ----------------
+@pytest.mark.parametrize("metric", [orc.L2, orc.IP, orc.COSINE])
+def test_flat_parity(metric):
+    base, q = make_data()
+    n, d = base.shape
+    k = 10
+    idx = dg.Index(dg.FLAT, metric, d)
+    try:
+        idx.add(np.arange(n, dtype=np.int64), base)
+        gd, gi = idx.search(q, k)
+    finally:
+        idx.close()
+    ob = base.copy()
+    oq = q.copy()
+    if metric == orc.COSINE:
+        ob = orc.normalize(ob)
+        oq = orc.normalize(oq)
+    od, oi = orc.flat_search(metric, ob, oq, k)
+    frac = ids_match_with_tie_slack(gd, gi, od, oi)
+    assert frac >= 0.99, f"metric {metric}: {frac}"
+
+
+def test_flat_self_top1():
+    # the reference's own pin (test_vector_index_recall_flat.cc:170-236)
+    base, _ = make_data(n=5000, d=64)
+    idx = dg.Index(dg.FLAT, dg.L2, 64)
+    try:
+        ids = np.arange(5000, dtype=np.int64) * 3 + 11
+        idx.add(ids, base)
+        gd, gi = idx.search(base, 1)
+        assert np.array_equal(gi[:, 0], ids)
+        assert np.all(np.abs(gd[:, 0]) < 1e-2)
+    finally:
+        idx.close()
+
+
+def test_flat_explicit_ids_padding_and_k_gt_n():
+    base, _ = make_data(n=5, d=32)
+    idx = dg.Index(dg.FLAT, dg.L2, 32)
+    try:
+        ids = np.array([100, 50, 7, 900, 3], np.int64)
+        idx.add(ids, base)
+        gd, gi = idx.search(base[:2], 8)
+        assert set(gi[0, :5]) == set(ids.tolist())
+        assert (gi[:, 5:] == -1).all()
+        assert (gd[:, 5:] == 0).all()
+    finally:
+        idx.close()
+
+
+def test_empty_index_and_k0():
+    idx = dg.Index(dg.FLAT, dg.L2, 32)
+    try:
+        gd, gi = idx.search(np.zeros((3, 32), np.float32), 4)
+        assert (gi == -1).all()
+    finally:
+        idx.close()
+
+
+# ---------------- IVF-Flat ----------------
+def build_pair(metric, base, nlist, seed=1234):
+    """GPU index + oracle structure sharing centroids."""
+    n, d = base.shape
+    ob = base.copy()
+    if metric == orc.COSINE:
+        ob = orc.normalize(ob)
+    cents = orc.kmeans(metric, ob, nlist, seed=seed)
+    gpu = dg.Index(dg.IVF_FLAT, metric, d, nlist=nlist)
+    gpu.set_centroids(cents)
+    gpu.add(np.arange(n, dtype=np.int64), base)
+    assign = orc.ivf_assign(metric, ob, cents)
+    off, gv, gi = orc.ivf_build(ob, None, nlist, assign)
+    return gpu, (cents, off, gv, gi)
+
+
+@pytest.mark.parametrize("metric", [orc.L2, orc.IP, orc.COSINE])
+def test_ivf_parity(metric):
+    base, q = make_data(n=30000, d=128, nq=128)
+    nlist, nprobe, k = 64, 16, 10
+    gpu, (cents, off, gv, gi_) = build_pair(metric, base, nlist)
+    try:
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    oq = q.copy()
+    if metric == orc.COSINE:
+        oq = orc.normalize(oq)
+    od, oi = orc.ivf_search(metric, cents, off, gv, gi_, oq, k, nprobe)
+    frac = ids_match_with_tie_slack(gd, gi, od, oi)
+    assert frac >= 0.98, f"metric {metric}: {frac}"
+
+
+def test_ivf_recall_equals_oracle():
+    base, q = make_data(n=30000, d=128, nq=256)
+    nlist, nprobe, k = 128, 8, 10
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_, q, k, nprobe)
+    gt_d, gt_i = orc.flat_search(orc.L2, base, q, k)
+    r_gpu = np.mean([len(set(a) & set(b)) / k for a, b in zip(gt_i, gi)])
+    r_orc = np.mean([len(set(a) & set(b)) / k for a, b in zip(gt_i, oi)])
+    assert abs(r_gpu - r_orc) <= 0.001, (r_gpu, r_orc)  # ±0.1pp, §8c
+
+
+def test_ivf_nprobe_default_and_clamp():
+    # nprobe<=0 -> default 80 clamped to nlist (ivf_flat.cc:208-214,234)
+    base, q = make_data(n=5000, d=64, nq=16)
+    nlist, k = 16, 5
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        gd0, gi0 = gpu.search(q, k, nprobe=0)       # default 80 -> clamp 16
+        gd1, gi1 = gpu.search(q, k, nprobe=9999)    # clamp 16
+    finally:
+        gpu.close()
+    od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_, q, k, nlist)
+    assert ids_match_with_tie_slack(gd0, gi0, od, oi) >= 0.99
+    assert ids_match_with_tie_slack(gd1, gi1, od, oi) >= 0.99
+
+
+def test_ivf_untrained_returns_blank():
+    idx = dg.Index(dg.IVF_FLAT, dg.L2, 64, nlist=16)
+    try:
+        gd, gi = idx.search(np.zeros((4, 64), np.float32), 3)
+        assert (gi == -1).all()  # ivf_flat.cc:223-227 "direct return blank"
+    finally:
+        idx.close()
+
+
+def test_ivf_add_before_train_fails():
+    idx = dg.Index(dg.IVF_FLAT, dg.L2, 64, nlist=16)
+    try:
+        with pytest.raises(dg.DgError) as e:
+            idx.add(np.arange(5, dtype=np.int64),
+                    np.zeros((5, 64), np.float32))
+        assert "NOT_TRAIN" in str(e.value) or "status 2" in str(e.value)
+    finally:
+        idx.close()
+
+
+def test_gpu_train_recall_close_to_oracle_train():
+    """GPU k-means need not be bit-identical; recall-equivalence judges it
+    (SURVEY.md §7 hard-part b)."""
+    base, q = make_data(n=30000, d=64, nq=256)
+    nlist, nprobe, k = 64, 8, 10
+    gt_d, gt_i = orc.flat_search(orc.L2, base, q, k)
+    gpu = dg.Index(dg.IVF_FLAT, dg.L2, 64, nlist=nlist)
+    try:
+        gpu.train(base)
+        gpu.add(np.arange(base.shape[0], dtype=np.int64), base)
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    cents = orc.kmeans(orc.L2, base, nlist)
+    assign = orc.ivf_assign(orc.L2, base, cents)
+    off, gv, gi_ = orc.ivf_build(base, None, nlist, assign)
+    od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_, q, k, nprobe)
+    r_gpu = np.mean([len(set(a) & set(b)) / k for a, b in zip(gt_i, gi)])
+    r_orc = np.mean([len(set(a) & set(b)) / k for a, b in zip(gt_i, oi)])
+    assert r_gpu >= r_orc - 0.02, (r_gpu, r_orc)
+
+
+# ---------------- filters ----------------
+def test_filter_range():
+    base, q = make_data(n=10000, d=64, nq=32)
+    idx = dg.Index(dg.FLAT, dg.L2, 64)
+    try:
+        idx.add(np.arange(10000, dtype=np.int64), base)
+        f = dg.make_filter(kind=1, min_id=1000, max_id=2000)
+        gd, gi = idx.search(q, 10, filt=f)
+    finally:
+        idx.close()
+    valid = gi[gi >= 0]
+    assert ((valid >= 1000) & (valid < 2000)).all()
+    od, oi = orc.flat_search(orc.L2, base[1000:2000], q, 10,
+                             ids=np.arange(1000, 2000, dtype=np.int64))
+    assert ids_match_with_tie_slack(gd, gi, od, oi) >= 0.99
+
+
+def test_filter_sorted_ids_and_negate():
+    base, q = make_data(n=5000, d=64, nq=16)
+    keep = np.sort(rng.choice(5000, 500, replace=False)).astype(np.int64)
+    idx = dg.Index(dg.FLAT, dg.L2, 64)
+    try:
+        idx.add(np.arange(5000, dtype=np.int64), base)
+        f = dg.make_filter(kind=2, ids=keep)
+        gd, gi = idx.search(q, 10, filt=f)
+        valid = gi[gi >= 0]
+        assert np.isin(valid, keep).all()
+        fneg = dg.make_filter(kind=2, ids=keep, negate=True)
+        gdn, gin = idx.search(q, 10, filt=fneg)
+        validn = gin[gin >= 0]
+        assert not np.isin(validn, keep).any()
+    finally:
+        idx.close()
+
+
+def test_filter_ivf_range():
+    base, q = make_data(n=20000, d=64, nq=32)
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, 32)
+    try:
+        f = dg.make_filter(kind=1, min_id=5000, max_id=15000)
+        gd, gi = gpu.search(q, 10, nprobe=32, filt=f)  # all lists
+    finally:
+        gpu.close()
+    valid = gi[gi >= 0]
+    assert ((valid >= 5000) & (valid < 15000)).all()
+    od, oi = orc.flat_search(orc.L2, base[5000:15000], q, 10,
+                             ids=np.arange(5000, 15000, dtype=np.int64))
+    assert ids_match_with_tie_slack(gd, gi, od, oi) >= 0.99
+
+
+# ---------------- mutation ----------------
+def test_remove_and_upsert():
+    base, q = make_data(n=2000, d=64, nq=8)
+    idx = dg.Index(dg.FLAT, dg.L2, 64)
+    try:
+        ids = np.arange(2000, dtype=np.int64)
+        idx.add(ids, base)
+        gd, gi = idx.search(base[:8], 1)
+        assert np.array_equal(gi[:, 0], ids[:8])
+        idx.remove(ids[:8])
+        gd, gi = idx.search(base[:8], 1)
+        assert not np.isin(gi[:, 0], ids[:8]).any()
+        # remove missing id -> error, nothing removed
+        with pytest.raises(dg.DgError):
+            idx.remove(np.array([0], np.int64))
+        # upsert puts them back (new vectors)
+        idx.upsert(ids[:8], base[8:16])
+        gd, gi = idx.search(base[8:16], 1)
+        assert np.isin(gi[:, 0], np.concatenate([ids[:8], ids[8:16]])).all()
+    finally:
+        idx.close()
+
+
+def test_duplicate_add_rejected():
+    idx = dg.Index(dg.FLAT, dg.L2, 32)
+    try:
+        v = np.zeros((2, 32), np.float32)
+        idx.add(np.array([5, 6], np.int64), v)
+        with pytest.raises(dg.DgError) as e:
+            idx.add(np.array([6, 7], np.int64), v)
+        assert "duplicated" in str(e.value)
+    finally:
+        idx.close()
+
+
+# ---------------- save / load ----------------
+def test_save_load_roundtrip(tmp_path):
+    base, q = make_data(n=10000, d=64, nq=32)
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, 32)
+    p = str(tmp_path / "idx.dgi")
+    try:
+        gd0, gi0 = gpu.search(q, 10, nprobe=8)
+        gpu.save(p)
+    finally:
+        gpu.close()
+    idx2 = dg.Index.load(p)
+    try:
+        gd1, gi1 = idx2.search(q, 10, nprobe=8)
+    finally:
+        idx2.close()
+    assert np.array_equal(gi0, gi1)
+    assert np.allclose(gd0, gd1, rtol=1e-6, atol=1e-6)
+
+
+# ---------------- multi-shard invariance (single GPU, two indexes) ----------
+def test_row_shard_union_invariance_gpu():
+    base, q = make_data(n=20000, d=64, nq=64)
+    nlist, nprobe, k = 32, 8, 10
+    cents = orc.kmeans(orc.L2, base, nlist)
+    whole = dg.Index(dg.IVF_FLAT, dg.L2, 64, nlist=nlist)
+    whole.set_centroids(cents)
+    whole.add(np.arange(20000, dtype=np.int64), base)
+    try:
+        wd, wi = whole.search(q, k, nprobe=nprobe)
+    finally:
+        whole.close()
+    shard_d, shard_i = [], []
+    for r in range(2):
+        sl = slice(r * 10000, (r + 1) * 10000)
+        sh = dg.Index(dg.IVF_FLAT, dg.L2, 64, nlist=nlist)
+        sh.set_centroids(cents)
+        sh.add(np.arange(sl.start, sl.stop, dtype=np.int64), base[sl])
+        try:
+            sd, si = sh.search(q, k, nprobe=nprobe)
+        finally:
+            sh.close()
+        shard_d.append(sd)
+        shard_i.append(si)
+    md, mi = dg.merge_topk(np.stack(shard_d), np.stack(shard_i), k)
+    frac = ids_match_with_tie_slack(md, mi, wd, wi)
+    assert frac >= 0.99, frac
+
+
+# ---------------- C++ plugin mirror ----------------
+def test_cpp_mirror_selftest():
+    import ctypes
+    lib = ctypes.CDLL(os.path.join(REPO, "dingo-store_amd",
+                                   "libdingo_gpu.so"))
+    rc = lib.dg_mirror_selftest()
+    assert rc == 0, f"dg_mirror_selftest returned {rc}"
+
+
+# ---------------- stats ----------------
+def test_stats_timing_populated():
+    base, q = make_data(n=20000, d=128, nq=64)
+    gpu, _ = build_pair(orc.L2, base, 64)
+    try:
+        gpu.search(q, 10, nprobe=16)
+        st = gpu.stats()
+    finally:
+        gpu.close()
+    assert st["ntotal"] == 20000
+    assert st["last_nq"] == 64
+    assert st["last_scan_ms"] > 0
+    assert st["last_scan_bytes_algorithmic"] > 0
