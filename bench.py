@@ -1,0 +1,371 @@
+#!/usr/bin/env python3
+"""bench.py — the driver-contract benchmark for the MI355X-native
+dingo-store vector-search path.
+
+Measures BASELINE.json's metric — QPS @ recall@10 for IVF-Flat 10M x 768
+fp32, nlist=4096, nprobe=32, batch=1024, k=10 (cfg C; `config.workload`) —
+through the product C-ABI (dg_search_device), queries resident in HBM when
+the timed region starts.
+
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched by the driver via torch.distributed.run, one rank per GPU
+over RCCL: the database is row-sharded across ranks with replicated
+centroids; each step is local IVF search + RCCL all-gather of per-rank
+top-k over xGMI + device-side merge.  scaling is
+"strong": total work (10M rows x 1024 queries) is fixed as N grows.
+
+Data: synthetic, generated on-device per chunk with seeded torch
+generators (protocol in DESIGN.md §data; identical across ranks and across
+the GPU/CPU legs of one run).  The CPU baseline (rank 0, N=1) times the
+oracle's vectorized IVF search on the same structure/host cores — the
+checker timed as a baseline, never the product path.
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path[:0] = [REPO, os.path.join(REPO, "dingo-store_amd")]
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+import dingostore as dg  # noqa: E402
+
+GEN_CHUNK = 1 << 20  # rows per generated chunk
+
+
+def log(rank, *a):
+    if rank == 0:
+        print("[bench]", *a, file=sys.stderr, flush=True)
+
+
+def gen_chunk_device(seed, c, rows, d, device):
+    """Chunk c of the base set: fp32 uniform[0,1), torch philox on device.
+    (bench data protocol — DESIGN.md §data; deterministic per (seed, c))"""
+    g = torch.Generator(device=device)
+    g.manual_seed(seed * 1_000_003 + c)
+    return torch.rand((rows, d), generator=g, device=device,
+                      dtype=torch.float32)
+
+
+def gen_queries_device(seed, n, d, nq, device):
+    """Queries = base rows 0..nq-1 + N(0, 0.05) noise."""
+    assert nq <= GEN_CHUNK
+    base0 = gen_chunk_device(seed, 0, min(n, GEN_CHUNK), d, device)
+    g = torch.Generator(device=device)
+    g.manual_seed(seed * 1_000_003 + 777_777)
+    idx = torch.arange(nq, device=device) % base0.shape[0]
+    q = base0[idx] + 0.05 * torch.randn((nq, d), generator=g, device=device)
+    del base0
+    return q.contiguous()
+
+
+def build_index(args, rank, world, device):
+    """Create + train + populate this rank's shard.  Returns (index,
+    row_range)."""
+    n, d, nlist = args.n, args.d, args.nlist
+    r0 = rank * n // world
+    r1 = (rank + 1) * n // world
+    idx = dg.Index(dg.IVF_FLAT, dg.L2, d, nlist=nlist, device=device.index)
+    # reserve capacity to avoid grow-copies
+    # (Index ctor has no reserve param; pre-reserve via desc would need it —
+    #  growth is amortized 1.5x, acceptable)
+
+    # ---- train on rank 0 (first 256*nlist rows of the global set), then
+    # broadcast centroids (mirrors TrainForBuild + snapshot install roles)
+    t0 = time.time()
+    n_train = min(n, 256 * nlist)
+    if rank == 0:
+        chunks = []
+        got = 0
+        c = 0
+        while got < n_train:
+            rows = min(GEN_CHUNK, n - c * GEN_CHUNK, n_train - got)
+            chunks.append(gen_chunk_device(args.seed, c, rows, d, device))
+            got += rows
+            c += 1
+        train = torch.cat(chunks) if len(chunks) > 1 else chunks[0]
+        del chunks
+        idx.train(train.cpu().numpy())  # dg_train runs k-means on the GPU
+        del train
+        cents = torch.from_numpy(idx.get_centroids()).to(device)
+    else:
+        cents = torch.empty((nlist, d), device=device)
+    if world > 1:
+        torch.distributed.broadcast(cents, src=0)
+    if rank != 0:
+        idx.set_centroids(cents.cpu().numpy())
+    del cents
+    log(rank, f"train {time.time()-t0:.1f}s")
+
+    # ---- add this rank's rows chunk by chunk (generated on device,
+    # ingested via dg_add_device; assignment runs on the GPU)
+    t0 = time.time()
+    pos = r0
+    while pos < r1:
+        c = pos // GEN_CHUNK
+        c_start = c * GEN_CHUNK
+        rows_all = min(GEN_CHUNK, n - c_start)
+        lo, hi = pos - c_start, min(rows_all, r1 - c_start)
+        chunk = gen_chunk_device(args.seed, c, rows_all, d, device)
+        part = chunk[lo:hi].contiguous()
+        ids = np.arange(c_start + lo, c_start + hi, dtype=np.int64)
+        idx.add_device(ids, part.data_ptr(), part.shape[0])
+        del chunk, part
+        pos = c_start + hi
+    torch.cuda.synchronize()
+    log(rank, f"add {r1-r0} rows {time.time()-t0:.1f}s")
+    return idx, (r0, r1)
+
+
+def merged_step(idx, q, k, nprobe, dist_t, ids_t, world, gather_bufs):
+    """One timed step: local search + (N>1) RCCL all-gather + device merge.
+    Returns (dist, ids) tensors [nq, k] on device."""
+    idx.search_device(q.data_ptr(), q.shape[0], k, nprobe,
+                      dist_t.data_ptr(), ids_t.data_ptr())
+    idx.sync()
+    if world == 1:
+        return dist_t, ids_t
+    gd, gi = gather_bufs
+    torch.distributed.all_gather(gd, dist_t)
+    torch.distributed.all_gather(gi, ids_t)
+    cat_d = torch.cat(gd, dim=1)  # [nq, world*k]
+    cat_i = torch.cat(gi, dim=1)
+    key = torch.where(cat_i < 0, torch.full_like(cat_d, float("inf")), cat_d)
+    top = torch.topk(key, k, dim=1, largest=False)
+    return (torch.gather(cat_d, 1, top.indices),
+            torch.gather(cat_i, 1, top.indices))
+
+
+def compute_recall(idx, q, k, nprobe, nlist, world, device):
+    """recall@k of nprobe search vs exact (nprobe=nlist full sweep through
+    the same engine) — identical procedure for every N."""
+    nq = q.shape[0]
+    dist_t = torch.empty((nq, k), dtype=torch.float32, device=device)
+    ids_t = torch.empty((nq, k), dtype=torch.int64, device=device)
+    bufs = None
+    if world > 1:
+        bufs = ([torch.empty_like(dist_t) for _ in range(world)],
+                [torch.empty_like(ids_t) for _ in range(world)])
+    _, approx = merged_step(idx, q, k, nprobe, dist_t, ids_t, world, bufs)
+    approx = approx.clone()
+    _, exact = merged_step(idx, q, k, nlist, dist_t, ids_t, world, bufs)
+    hits = 0
+    a = approx.cpu().numpy()
+    e = exact.cpu().numpy()
+    for r in range(nq):
+        hits += len(set(a[r]) & set(e[r]))
+    return hits / (nq * k)
+
+
+def cpu_baseline(idx, args, q_host):
+    """Oracle (kind 'port') timed on host cores, same structure: centroids
+    from the GPU index, member lists from its assignments, base regenerated
+    on host from the same protocol. Bounded sample (~10-30 s)."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import pyoracle as orc
+    import psutil
+
+    n, d, nlist, nprobe, k = args.n, args.d, args.nlist, args.nprobe, args.k
+    avail = psutil.virtual_memory().available
+    need = n * d * 4 + (4 << 30)
+    frac = 1.0
+    if need > avail:
+        frac = max(0.02, (avail - (4 << 30)) / (n * d * 4))
+    n_cpu = int(n * frac)
+    # regenerate base rows [0, n_cpu) on host from the device protocol
+    base = np.empty((n_cpu, d), np.float32)
+    dev = torch.device("cuda:0")
+    for c in range((n_cpu + GEN_CHUNK - 1) // GEN_CHUNK):
+        rows = min(GEN_CHUNK, n - c * GEN_CHUNK)
+        t = gen_chunk_device(args.seed, c, rows, d, dev)
+        take = min(rows, n_cpu - c * GEN_CHUNK)
+        base[c * GEN_CHUNK:c * GEN_CHUNK + take] = t[:take].cpu().numpy()
+        del t
+    cents = idx.get_centroids()
+    assign = idx.export_assign()[:n_cpu]
+    order = np.argsort(assign, kind="stable")
+    member_rows = order.astype(np.int64)
+    counts = np.bincount(assign, minlength=nlist)
+    offsets = np.zeros(nlist + 1, np.int64)
+    np.cumsum(counts, out=offsets[1:])
+    # bounded sample: enough queries for ~10-30 s
+    t0 = time.time()
+    nq_probe = 8
+    orc.ivf_search_indexed_fast(orc.L2, cents, offsets, member_rows, base,
+                                q_host[:nq_probe], k, nprobe)
+    per_q = (time.time() - t0) / nq_probe
+    nq_sample = int(min(len(q_host), max(8, 20.0 / max(per_q, 1e-6))))
+    t0 = time.time()
+    orc.ivf_search_indexed_fast(orc.L2, cents, offsets, member_rows, base,
+                                q_host[:nq_sample], k, nprobe)
+    dt = time.time() - t0
+    cores = os.cpu_count()
+    sample = (f"{nq_sample} queries over "
+              f"{'full' if frac == 1.0 else f'{frac:.0%}-row-subsampled'} "
+              f"database ({n_cpu} rows), {dt:.1f}s, OpenMP {cores} cores")
+    return {"value": round(nq_sample / dt, 2), "unit": "queries/s",
+            "cores": cores, "kind": "port", "sample": sample}
+
+
+def read_traffic(workload_name):
+    """Per-launch HBM traffic measured by a separate rocprofv3 --pmc run
+    (profiles/pmc_traffic.json, written by profiling scripts); null if no
+    matching measurement exists."""
+    for p in (os.path.join(REPO, "gpurun_out", "pmc_traffic.json"),
+              os.path.join(REPO, "profiles", "pmc_traffic.json")):
+        try:
+            j = json.load(open(p))
+            if j.get("workload") == workload_name:
+                return j.get("bytes_per_launch")
+        except Exception:
+            pass
+    return None
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=1024)
+    ap.add_argument("--n", type=int, default=10_000_000)
+    ap.add_argument("--d", type=int, default=768)
+    ap.add_argument("--nlist", type=int, default=4096)
+    ap.add_argument("--nprobe", type=int, default=32)
+    ap.add_argument("--k", type=int, default=10)
+    ap.add_argument("--seed", type=int, default=4244)
+    ap.add_argument("--quick", action="store_true",
+                    help="reduced size for smoke runs (1M rows)")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-recall", action="store_true")
+    args = ap.parse_args()
+    if args.quick:
+        args.n, args.nlist = 1_000_000, 1024
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        torch.distributed.init_process_group("nccl")  # = RCCL on ROCm
+    assert torch.cuda.is_available(), "bench needs a GPU (no CPU fallback)"
+    torch.cuda.set_device(local_rank)
+    device = torch.device(f"cuda:{local_rank}")
+
+    idx, (r0, r1) = build_index(args, rank, world, device)
+    nq, k, nprobe = args.batch, args.k, args.nprobe
+    q = gen_queries_device(args.seed, args.n, args.d, nq, device)
+    dist_t = torch.empty((nq, k), dtype=torch.float32, device=device)
+    ids_t = torch.empty((nq, k), dtype=torch.int64, device=device)
+    bufs = None
+    if world > 1:
+        bufs = ([torch.empty_like(dist_t) for _ in range(world)],
+                [torch.empty_like(ids_t) for _ in range(world)])
+
+    # ---- warmup (includes CSR finalize on first search)
+    t0 = time.time()
+    for _ in range(args.warmup):
+        merged_step(idx, q, k, nprobe, dist_t, ids_t, world, bufs)
+    torch.cuda.synchronize()
+    log(rank, f"warmup {time.time()-t0:.1f}s")
+
+    # ---- timed region
+    if world > 1:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        merged_step(idx, q, k, nprobe, dist_t, ids_t, world, bufs)
+    idx.sync()
+    torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    elapsed = time.time() - t0
+    if world > 1:
+        e = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    # ---- per-stage stats from extra untimed steps (hipEvent timings on the
+    # library stream; torch events would miss it)
+    scan_ms, coarse_ms, total_ms, alg_bytes = [], [], [], []
+    for _ in range(5):
+        merged_step(idx, q, k, nprobe, dist_t, ids_t, world, bufs)
+        st = idx.stats()
+        scan_ms.append(st["last_scan_ms"])
+        coarse_ms.append(st["last_coarse_ms"])
+        total_ms.append(st["last_total_ms"])
+        alg_bytes.append(st["last_scan_bytes_algorithmic"])
+
+    recall = None
+    if not args.no_recall:
+        recall = compute_recall(idx, q, k, nprobe, args.nlist, world, device)
+
+    cpu = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        q_host = q.cpu().numpy()
+        try:
+            cpu = cpu_baseline(idx, args, q_host)
+        except Exception as ex:
+            log(rank, f"cpu baseline failed: {ex}")
+
+    if rank == 0:
+        qps = nq * args.steps / elapsed
+        scan_s = float(np.mean(scan_ms))
+        alg = int(np.mean(alg_bytes))
+        peak_gbps = 8000.0  # HBM3E spec (measured ceiling ~6300 GB/s,
+        # MI355X_MICROARCH.md); fraction vs spec per §8d
+        achieved = alg / (scan_s * 1e6) if scan_s > 0 else 0.0
+        workload_name = (f"IVF-Flat {args.n//10**6}M x {args.d} fp32 "
+                         f"nlist={args.nlist} nprobe={nprobe} "
+                         f"batch={nq} k={k}")
+        out = {
+            "metric": "QPS @ recall@10, IVF-Flat 10Mx768 nprobe=32",
+            "value": round(qps, 1),
+            "unit": "queries/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,  # no published reference numbers
+            "dtype": "f32",
+            "data": "synthetic (seeded torch philox uniform[0,1); queries = "
+                    "base + N(0,0.05); DESIGN.md §data)",
+            "recall_at_k": round(recall, 4) if recall is not None else None,
+            "config": {
+                "workload": workload_name,
+                "n": args.n, "d": args.d, "nlist": args.nlist,
+                "nprobe": nprobe, "batch": nq, "k": k,
+                "parallelism": f"row-sharded ivf x{world}, RCCL all-gather "
+                               "top-k" if world > 1 else "single GPU",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": round(achieved, 1),
+                "peak": peak_gbps,
+                "unit": "GB/s",
+                "frac": round(achieved / peak_gbps, 4),
+                "traffic": read_traffic(workload_name),
+                "detail": {
+                    "scan_ms_per_launch": round(scan_s, 3),
+                    "coarse_ms": round(float(np.mean(coarse_ms)), 3),
+                    "total_ms": round(float(np.mean(total_ms)), 3),
+                    "algorithmic_bytes_per_launch": alg,
+                },
+            },
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(out), flush=True)
+    idx.close()
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

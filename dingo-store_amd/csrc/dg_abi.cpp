@@ -180,6 +180,17 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
   for (auto& e : ix->ev) (void)hipEventCreate(&e);
   ix->events_ready = true;
   if (desc.kind == DG_INDEX_FLAT) ix->trained = true;
+  if (desc.reserve > 0) {  // capacity hint: avoids grow-copies during add
+    if (dbuf_reserve(ix->d_vectors, (size_t)desc.reserve * desc.d * 4,
+                     ix->stream, false) != DG_OK ||
+        dbuf_reserve(ix->d_ids, (size_t)desc.reserve * 8, ix->stream,
+                     false) != DG_OK ||
+        dbuf_reserve(ix->d_assign, (size_t)desc.reserve * 4, ix->stream,
+                     false) != DG_OK) {
+      dg_index_destroy(ix);
+      return DG_ENOMEM;
+    }
+  }
   *out = ix;
   return DG_OK;
 }
@@ -431,8 +442,8 @@ extern "C" dg_status dg_train(dg_index* ix, int64_t n, const float* x) {
 }
 
 // ---------------- add / remove ----------------
-extern "C" dg_status dg_add(dg_index* ix, int64_t n, const int64_t* ids,
-                            const float* x) {
+static dg_status add_impl(dg_index* ix, int64_t n, const int64_t* ids,
+                          const float* x, bool x_on_device) {
   if (!ix || !ids || !x || n <= 0) {
     dg_set_error("bad add args");
     return DG_EINVAL;
@@ -474,8 +485,10 @@ extern "C" dg_status dg_add(dg_index* ix, int64_t n, const int64_t* ids,
     st = dbuf_reserve(ix->d_assign, (size_t)(n0 + n) * 4, ix->stream, true);
   if (st != DG_OK) return st;
   float* dst = (float*)ix->d_vectors.p + (size_t)n0 * d;
-  DG_HIP_CHECK(hipMemcpyAsync(dst, x, (size_t)n * d * 4,
-                              hipMemcpyHostToDevice, ix->stream));
+  DG_HIP_CHECK(hipMemcpyAsync(
+      dst, x, (size_t)n * d * 4,
+      x_on_device ? hipMemcpyDeviceToDevice : hipMemcpyHostToDevice,
+      ix->stream));
   DG_HIP_CHECK(hipMemcpyAsync((int64_t*)ix->d_ids.p + n0, ids, (size_t)n * 8,
                               hipMemcpyHostToDevice, ix->stream));
   if (ix->desc.metric == DG_METRIC_COSINE)
@@ -490,6 +503,26 @@ extern "C" dg_status dg_add(dg_index* ix, int64_t n, const int64_t* ids,
   for (int64_t i = 0; i < n; i++) ix->id_count.emplace(ids[i], 1);
   ix->ntotal += n;
   ix->csr_valid = false;
+  return DG_OK;
+}
+
+extern "C" dg_status dg_add(dg_index* ix, int64_t n, const int64_t* ids,
+                            const float* x) {
+  return add_impl(ix, n, ids, x, false);
+}
+
+extern "C" dg_status dg_add_device(dg_index* ix, int64_t n,
+                                   const int64_t* ids, const float* d_x) {
+  return add_impl(ix, n, ids, d_x, true);
+}
+
+extern "C" dg_status dg_export_assign(dg_index* ix, int32_t* out) {
+  if (!ix || !out) return DG_EINVAL;
+  std::shared_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  if (ix->ntotal == 0) return DG_OK;
+  DG_HIP_CHECK(hipMemcpy(out, ix->d_assign.p, (size_t)ix->ntotal * 4,
+                         hipMemcpyDeviceToHost));
   return DG_OK;
 }
 
@@ -752,21 +785,33 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     uint64_t* coarse_tk = (uint64_t*)ix->ws_probes.p;        // nq x np
     int32_t* probes = (int32_t*)(coarse_tk + (size_t)nq * np);  // nq x np
     uint64_t* final_tk = (uint64_t*)(probes + (size_t)nq * np);  // nq x k
-    st = sgemm_dots(ix, dq, nq, (const float*)ix->d_centroids.p, nlist, d,
-                    (float*)ix->ws_dots.p);
-    if (st != DG_OK) {
-      dbuf_free(d_fids);
-      dbuf_free(ws_bitmap);
-      return st;
+    const uint8_t* maskp =
+        ix->has_mask ? (const uint8_t*)ix->d_list_mask.p : nullptr;
+    if (np == nlist) {
+      // full sweep: every list probed; no coarse selection needed (also the
+      // exact-ground-truth path recall measurement uses)
+      dgk::probes_all(ix->stream, nq, np, maskp, probes);
+    } else {
+      if (np > 128) {
+        dg_set_error("nprobe %d > 128 (and < nlist) unsupported this round",
+                     np);
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        return DG_ENOT_SUPPORT;
+      }
+      st = sgemm_dots(ix, dq, nq, (const float*)ix->d_centroids.p, nlist, d,
+                      (float*)ix->ws_dots.p);
+      if (st != DG_OK) {
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        return st;
+      }
+      dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
+                        (const float*)ix->d_cnorms.p, nq, nlist, np,
+                        metric == DG_METRIC_L2 ? 1 : 2, nullptr, 0, coarse_tk,
+                        np, 0);
+      dgk::probe_unpack(ix->stream, coarse_tk, nq, np, maskp, probes);
     }
-    dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
-                      (const float*)ix->d_cnorms.p, nq, nlist, np,
-                      metric == DG_METRIC_L2 ? 1 : 2, nullptr, 0, coarse_tk,
-                      np, 0);
-    dgk::probe_unpack(ix->stream, coarse_tk, nq, np,
-                      ix->has_mask ? (const uint8_t*)ix->d_list_mask.p
-                                   : nullptr,
-                      probes);
     (void)hipEventRecord(ix->ev[1], ix->stream);
 
     // inverted mapping + candidate offsets

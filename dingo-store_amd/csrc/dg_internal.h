@@ -116,6 +116,8 @@ struct dg_index {
 namespace dgk {
 void probe_unpack(hipStream_t s, const uint64_t* topk, int64_t nq,
                   int32_t nprobe, const uint8_t* mask, int32_t* probes);
+void probes_all(hipStream_t s, int64_t nq, int32_t nprobe,
+                const uint8_t* mask, int32_t* probes);
 void init_cursors(hipStream_t s, const int64_t* offsets, int32_t n,
                   int32_t* cursors);
 void select_dense(hipStream_t s, const float* scores, const float* cnorms,

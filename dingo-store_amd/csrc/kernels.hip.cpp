@@ -285,6 +285,16 @@ __global__ void k_probe_unpack(const uint64_t* __restrict__ topk, int64_t nq,
   probes[i] = l;
 }
 
+__global__ void k_probes_all(int64_t nq, int32_t nprobe, const uint8_t* mask,
+                             int32_t* __restrict__ probes) {
+  // nprobe == nlist: probe every list in id order (no coarse selection)
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= nq * nprobe) return;
+  int32_t l = (int32_t)(i % nprobe);
+  if (mask && !mask[l]) l = -1;
+  probes[i] = l;
+}
+
 __global__ void k_hist_probes(const int32_t* __restrict__ probes, int64_t n,
                               int32_t* __restrict__ counts) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -652,6 +662,12 @@ void emit_results(hipStream_t s, const uint64_t* topk,
   hipLaunchKernelGGL(k_emit, dim3(ceil_div(nq * k, 256)), dim3(256), 0, s,
                      topk, ids_lookup, qnorms, nq, k, metric, add_qnorm,
                      out_dist, out_ids);
+}
+
+void probes_all(hipStream_t s, int64_t nq, int32_t nprobe,
+                const uint8_t* mask, int32_t* probes) {
+  hipLaunchKernelGGL(k_probes_all, dim3(ceil_div(nq * nprobe, 256)),
+                     dim3(256), 0, s, nq, nprobe, mask, probes);
 }
 
 void probe_unpack(hipStream_t s, const uint64_t* topk, int64_t nq,

@@ -183,6 +183,24 @@ class Index:
         ids = np.ascontiguousarray(ids, np.int64)
         _check(lib().dg_add(self.h, x.shape[0], ids, x), "dg_add")
 
+    def add_device(self, ids, x_ptr, n):
+        """Add n vectors from a device pointer (torch .data_ptr())."""
+        ids = np.ascontiguousarray(ids, np.int64)
+        l = lib()
+        l.dg_add_device.argtypes = [C.c_void_p, C.c_int64, _i64p, C.c_void_p]
+        _check(l.dg_add_device(self.h, n, ids, C.c_void_p(x_ptr)),
+               "dg_add_device")
+
+    def export_assign(self):
+        l = lib()
+        l.dg_export_assign.argtypes = [
+            C.c_void_p, np.ctypeslib.ndpointer(np.int32,
+                                               flags="C_CONTIGUOUS")]
+        st = self.stats()
+        out = np.empty(st["ntotal"], np.int32)
+        _check(l.dg_export_assign(self.h, out), "dg_export_assign")
+        return out
+
     def upsert(self, ids, x):
         x = np.ascontiguousarray(x, np.float32)
         ids = np.ascontiguousarray(ids, np.int64)

@@ -261,11 +261,14 @@ extern "C" int dg_mirror_selftest(void) {
     if (res.size() != (size_t)n) return 4;
     for (int i = 0; i < n; i++) {
       if (res[i].vector_with_distances.empty()) return 5;
-      // self-top-1 (test_vector_index_recall_flat.cc:170-236)
-      if (res[i].vector_with_distances[0].vector_with_id.id != batch[i].id)
+      // self-top-1 (test_vector_index_recall_flat.cc:170-236) holds for L2
+      // and cosine; NOT for raw inner product (the max-IP neighbor of x
+      // need not be x itself)
+      if (m != MetricType::kInnerProduct &&
+          res[i].vector_with_distances[0].vector_with_id.id != batch[i].id)
         return 6;
       float dist = res[i].vector_with_distances[0].distance;
-      // L2 self-dist 0; IP/cos reported as 1 - score
+      // L2 self-dist 0; cosine reported as 1 - score -> 0 for self
       if (m == MetricType::kL2 && std::fabs(dist) > 1e-3f) return 7;
       if (m == MetricType::kCosine && std::fabs(dist) > 1e-3f) return 8;
     }

@@ -151,6 +151,14 @@ dg_status dg_get_centroids(dg_index* idx, float* out_centroids);
  * DG_EID_DUPLICATED (src/vector/vector_index_utils.cc CheckVectorIdDuplicated).
  * upsert = remove-if-present + add (VectorIndexFlat::Upsert semantics). */
 dg_status dg_add(dg_index* idx, int64_t n, const int64_t* ids, const float* x);
+/* device-pointer add: d_x lives on the index's device (bench hot-path
+ * ingestion; semantics identical to dg_add) */
+dg_status dg_add_device(dg_index* idx, int64_t n, const int64_t* ids,
+                        const float* d_x);
+/* download the per-vector coarse assignment (arrival order, n = ntotal
+ * including tombstones).  Lets the CPU-baseline leg rebuild the identical
+ * IVF structure without re-running assignment on host. */
+dg_status dg_export_assign(dg_index* idx, int32_t* out /* ntotal */);
 dg_status dg_upsert(dg_index* idx, int64_t n, const int64_t* ids, const float* x);
 /* remove: all ids must exist, else DG_ENOT_FOUND and nothing is removed
  * (src/vector/vector_index_ivf_flat.cc:177-186). */

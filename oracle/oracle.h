@@ -114,6 +114,14 @@ void dgo_ivfpq_search(int metric, int32_t nlist, int32_t d,
 void dgo_flat_search_fast(int metric, int64_t n, int32_t d, const float* base,
                           const int64_t* ids, int64_t nq, const float* queries,
                           int32_t k, float* out_dist, int64_t* out_ids);
+void dgo_ivf_search_indexed_fast(int metric, int32_t nlist, int32_t d,
+                                 const float* centroids,
+                                 const int64_t* member_offsets,
+                                 const int64_t* member_rows,
+                                 const float* base, int64_t nq,
+                                 const float* queries, int32_t k,
+                                 int32_t nprobe, float* out_dist,
+                                 int64_t* out_ids);
 void dgo_ivf_search_fast(int metric, int32_t nlist, int32_t d,
                          const float* centroids, const int64_t* offsets,
                          const float* grouped_vectors,

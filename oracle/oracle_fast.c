@@ -104,6 +104,69 @@ void dgo_flat_search_fast(int metric, int64_t n, int32_t d, const float* base,
   }
 }
 
+void dgo_ivf_search_indexed_fast(int metric, int32_t nlist, int32_t d,
+                                 const float* centroids,
+                                 const int64_t* member_offsets,
+                                 const int64_t* member_rows,
+                                 const float* base /* ungrouped */,
+                                 int64_t nq, const float* queries, int32_t k,
+                                 int32_t nprobe, float* out_dist,
+                                 int64_t* out_ids) {
+  /* CPU-baseline variant that scans lists through a row-index indirection
+   * (member_rows grouped by list via member_offsets) over the UNGROUPED
+   * base array — same arithmetic, avoids a second 30 GB host copy at
+   * BASELINE cfg C (BASELINE.md / DESIGN.md §cpu-baseline). ids are the
+   * row indices themselves. */
+  if (nprobe > nlist) nprobe = nlist;
+#pragma omp parallel
+  {
+    fcand* probes = (fcand*)malloc(sizeof(fcand) * nlist);
+    fcand* heap = (fcand*)malloc(sizeof(fcand) * k);
+#pragma omp for schedule(dynamic, 1)
+    for (int64_t q = 0; q < nq; q++) {
+      const float* qv = queries + (size_t)q * d;
+      for (int32_t l = 0; l < nlist; l++) {
+        probes[l].key = key_fast(metric, qv, centroids + (size_t)l * d, d);
+        probes[l].id = l;
+      }
+      qsort(probes, nlist, sizeof(fcand), fcand_cmp);
+      int32_t size = 0, worst = 0;
+      for (int32_t p = 0; p < nprobe; p++) {
+        int32_t l = (int32_t)probes[p].id;
+        for (int64_t m = member_offsets[l]; m < member_offsets[l + 1]; m++) {
+          int64_t row = member_rows[m];
+          float key = key_fast(metric, qv, base + (size_t)row * d, d);
+          if (size < k) {
+            heap[size].key = key;
+            heap[size].id = row;
+            if (size == 0 || key > heap[worst].key) worst = size;
+            size++;
+          } else if (key < heap[worst].key) {
+            heap[worst].key = key;
+            heap[worst].id = row;
+            worst = 0;
+            for (int32_t j = 1; j < size; j++)
+              if (heap[j].key > heap[worst].key) worst = j;
+          }
+        }
+      }
+      qsort(heap, size, sizeof(fcand), fcand_cmp);
+      for (int32_t i = 0; i < k; i++) {
+        if (i < size) {
+          out_dist[(size_t)q * k + i] =
+              (metric == DGO_L2) ? heap[i].key : -heap[i].key;
+          out_ids[(size_t)q * k + i] = heap[i].id;
+        } else {
+          out_dist[(size_t)q * k + i] = 0.0f;
+          out_ids[(size_t)q * k + i] = -1;
+        }
+      }
+    }
+    free(probes);
+    free(heap);
+  }
+}
+
 void dgo_ivf_search_fast(int metric, int32_t nlist, int32_t d,
                          const float* centroids, const int64_t* offsets,
                          const float* grouped_vectors,

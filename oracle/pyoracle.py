@@ -157,6 +157,28 @@ def ivf_search(metric, centroids, offsets, gv, gi, queries, k, nprobe,
     return out_dist, out_ids
 
 
+def ivf_search_indexed_fast(metric, centroids, member_offsets, member_rows,
+                            base, queries, k, nprobe):
+    """CPU-baseline leg: indexed scan over the UNGROUPED base array."""
+    centroids = np.ascontiguousarray(centroids, np.float32)
+    queries = np.ascontiguousarray(queries, np.float32)
+    base = np.ascontiguousarray(base, np.float32)
+    nlist, d = centroids.shape
+    nq = queries.shape[0]
+    out_dist = np.empty((nq, k), np.float32)
+    out_ids = np.empty((nq, k), np.int64)
+    fn = _lib.dgo_ivf_search_indexed_fast
+    fn.argtypes = [
+        C.c_int, C.c_int32, C.c_int32, _f32p, _i64p, _i64p, _f32p, C.c_int64,
+        _f32p, C.c_int32, C.c_int32, _f32p, _i64p,
+    ]
+    fn(metric, nlist, d, centroids,
+       np.ascontiguousarray(member_offsets, np.int64),
+       np.ascontiguousarray(member_rows, np.int64), base, nq, queries, k,
+       nprobe, out_dist, out_ids)
+    return out_dist, out_ids
+
+
 def pq_train(residuals, m, nbits=8, seed=1234):
     residuals = np.ascontiguousarray(residuals, np.float32)
     n, d = residuals.shape
