@@ -122,6 +122,18 @@ def build_index(args, rank, world, device):
     return idx, (r0, r1)
 
 
+def torch_merge(gd, gi, k, metric_l2=True):
+    """Merge per-shard top-k lists (the RCCL all-gather consumer), device or
+    CPU tensors.  gd/gi: lists of [nq, k] tensors in faiss convention."""
+    cat_d = torch.cat(gd, dim=1)  # [nq, world*k]
+    cat_i = torch.cat(gi, dim=1)
+    key = cat_d if metric_l2 else -cat_d
+    key = torch.where(cat_i < 0, torch.full_like(key, float("inf")), key)
+    top = torch.topk(key, k, dim=1, largest=False)
+    return (torch.gather(cat_d, 1, top.indices),
+            torch.gather(cat_i, 1, top.indices))
+
+
 def merged_step(idx, q, k, nprobe, dist_t, ids_t, world, gather_bufs):
     """One timed step: local search + (N>1) RCCL all-gather + device merge.
     Returns (dist, ids) tensors [nq, k] on device."""
@@ -133,12 +145,7 @@ def merged_step(idx, q, k, nprobe, dist_t, ids_t, world, gather_bufs):
     gd, gi = gather_bufs
     torch.distributed.all_gather(gd, dist_t)
     torch.distributed.all_gather(gi, ids_t)
-    cat_d = torch.cat(gd, dim=1)  # [nq, world*k]
-    cat_i = torch.cat(gi, dim=1)
-    key = torch.where(cat_i < 0, torch.full_like(cat_d, float("inf")), cat_d)
-    top = torch.topk(key, k, dim=1, largest=False)
-    return (torch.gather(cat_d, 1, top.indices),
-            torch.gather(cat_i, 1, top.indices))
+    return torch_merge(gd, gi, k)
 
 
 def compute_recall(idx, q, k, nprobe, nlist, world, device):

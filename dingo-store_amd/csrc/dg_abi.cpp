@@ -162,6 +162,10 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
   }
   if (desc.kind == DG_INDEX_IVF_FLAT && desc.nlist <= 0)
     desc.nlist = 2048;  // kCreateIvfFlatParamNcentroids, constant.h:177
+  if (desc.kind == DG_INDEX_IVF_FLAT && desc.d > 2048) {
+    dg_set_error("IVF scan kernel supports d <= 2048 this round");
+    return DG_ENOT_SUPPORT;
+  }
   if (dg_device_count() == 0) {
     dg_set_error("no HIP device (the GPU path has no CPU fallback)");
     return DG_ENOGPU;

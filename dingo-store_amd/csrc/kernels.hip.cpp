@@ -362,6 +362,11 @@ __global__ void k_fill_units(const int32_t* __restrict__ unit_offsets,
 // probing the list are processed in LDS-staged tiles of QT.  Wave-per-row,
 // float4 coalesced; per (row, query) distance by wave shuffle reduction;
 // candidates written as packed u64 at precomputed dense offsets.
+// PER_LANE (= ceil(d/256)) is a template parameter so the row registers stay
+// in VGPRs (a runtime bound would spill the array to scratch) and the next
+// row is prefetched while the current one is reduced (2 rows in flight per
+// wave on top of wave-level parallelism).
+template <int PER_LANE>
 __global__ void __launch_bounds__(256, 2) k_ivf_scan(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const float* __restrict__ csr_vectors, const float* __restrict__ csr_vnorms,
@@ -402,43 +407,61 @@ __global__ void __launch_bounds__(256, 2) k_ivf_scan(
       int32_t j = threadIdx.x;
       int32_t q = inv_q[iq0 + t0 + j];
       int32_t rank = inv_rank[iq0 + t0 + j];
-      // cand position for row r = q's base + this probe's offset + (r - list_start)
-      cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] - list_start;
+      // cand position for row r = q's base + this probe's offset +
+      // (r - list_start)
+      cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                 list_start;
     }
     __syncthreads();
+    const int64_t my_cbase = (lane < qt) ? cbase[lane] : 0;
 
-    for (int64_t r = row_start + wave_id; r < row_end; r += nwaves) {
+    float4 cur[PER_LANE], nxt[PER_LANE];
+    int64_t r = row_start + wave_id;
+    if (r < row_end) {
       const float4* v4 = (const float4*)(csr_vectors + (size_t)r * d);
-      // load row into registers (up to 3 float4 per lane at d=768)
-      float4 reg[8];
-      const int per_lane = (d4 + WAVE - 1) / WAVE;
-      for (int i = 0; i < per_lane; i++) {
+#pragma unroll
+      for (int i = 0; i < PER_LANE; i++) {
         int idx = lane + i * WAVE;
-        reg[i] = (idx < d4) ? v4[idx]
-                            : make_float4(0.f, 0.f, 0.f, 0.f);
+        cur[i] = (idx < d4) ? v4[idx] : make_float4(0.f, 0.f, 0.f, 0.f);
+      }
+    }
+    for (; r < row_end; r += nwaves) {
+      const int64_t rn = r + nwaves;
+      if (rn < row_end) {  // prefetch next row while reducing this one
+        const float4* v4n = (const float4*)(csr_vectors + (size_t)rn * d);
+#pragma unroll
+        for (int i = 0; i < PER_LANE; i++) {
+          int idx = lane + i * WAVE;
+          nxt[i] = (idx < d4) ? v4n[idx] : make_float4(0.f, 0.f, 0.f, 0.f);
+        }
       }
       bool pass = true;
       if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
       const float vn = (metric == 0) ? csr_vnorms[r] : 0.f;
 
+      float mykey = 0.f;  // lane j ends up holding query j's key for row r
       for (int32_t j = 0; j < qt; j++) {
         const float4* q4 = (const float4*)(smem + (size_t)j * d);
         float acc = 0.f;
-        for (int i = 0; i < per_lane; i++) {
+#pragma unroll
+        for (int i = 0; i < PER_LANE; i++) {
           int idx = lane + i * WAVE;
           if (idx < d4) {
-            float4 a = reg[i], b = q4[idx];
+            float4 a = cur[i], b = q4[idx];
             acc += a.x * b.x + a.y * b.y + a.z * b.z + a.w * b.w;
           }
         }
         for (int off = 32; off; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
-        if (lane == 0) {
-          // L2 key = vnorm - 2*dot (+qnorm at emit); IP key = -dot
-          float key = (metric == 0) ? vn - 2.0f * acc : -acc;
-          uint64_t c = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
-          cand[cbase[j] + r] = c;
-        }
+        // L2 key = vnorm - 2*dot (+qnorm at emit); IP key = -dot
+        float key = (metric == 0) ? vn - 2.0f * acc : -acc;
+        if (lane == j) mykey = key;
       }
+      if (lane < qt) {  // one vector-store wave-instruction, qt active lanes
+        uint64_t c = pass ? pack_cand(mykey, (uint32_t)r) : kCandEmpty;
+        cand[my_cbase + r] = c;
+      }
+#pragma unroll
+      for (int i = 0; i < PER_LANE; i++) cur[i] = nxt[i];
     }
   }
 }
@@ -724,10 +747,28 @@ void ivf_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
               int32_t qt_max, uint64_t* cand) {
   if (!n_units) return;
   size_t lds = (size_t)qt_max * d * 4 + (size_t)qt_max * 8;
-  hipLaunchKernelGGL(k_ivf_scan, dim3((uint32_t)n_units), dim3(256), lds, s,
-                     units, csr_offsets, csr_vectors, csr_vnorms, queries, d,
-                     inv_offsets, inv_q, inv_rank, qp_off, q_cand_base,
-                     nprobe, metric, bitmap, chunk_rows, qt_max, cand);
+  const int per_lane = (d / 4 + WAVE - 1) / WAVE;  // ceil(d/256)
+#define DG_SCAN_CASE(PL)                                                     \
+  case PL:                                                                   \
+    hipLaunchKernelGGL(k_ivf_scan<PL>, dim3((uint32_t)n_units), dim3(256),   \
+                       lds, s, units, csr_offsets, csr_vectors, csr_vnorms,  \
+                       queries, d, inv_offsets, inv_q, inv_rank, qp_off,     \
+                       q_cand_base, nprobe, metric, bitmap, chunk_rows,      \
+                       qt_max, cand);                                        \
+    break;
+  switch (per_lane) {
+    DG_SCAN_CASE(1)
+    DG_SCAN_CASE(2)
+    DG_SCAN_CASE(3)
+    DG_SCAN_CASE(4)
+    DG_SCAN_CASE(5)
+    DG_SCAN_CASE(6)
+    DG_SCAN_CASE(7)
+    DG_SCAN_CASE(8)
+    default:
+      break;  // host validates d <= 2048 for IVF scan
+  }
+#undef DG_SCAN_CASE
 }
 
 void hist_assign(hipStream_t s, const int32_t* assign, int64_t n,
