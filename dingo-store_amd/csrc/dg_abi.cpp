@@ -206,7 +206,7 @@ extern "C" void dg_index_destroy(dg_index* ix) {
   for (auto b :
        {&ix->d_vectors, &ix->d_ids, &ix->d_assign, &ix->d_centroids,
         &ix->d_cnorms, &ix->d_csr_offsets, &ix->d_csr_vectors, &ix->d_csr_ids,
-        &ix->d_csr_vnorms, &ix->d_list_mask, &ix->ws_queries, &ix->ws_qnorms,
+        &ix->d_csr_vnorms, &ix->d_csr_t, &ix->d_chunk_meta, &ix->d_list_mask, &ix->ws_queries, &ix->ws_qnorms,
         &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
         &ix->ws_small, &ix->ws_topk})
     dbuf_free(*b);
@@ -606,23 +606,29 @@ extern "C" dg_status dg_set_list_mask(dg_index* ix, const uint8_t* mask) {
 
 // ---------------- finalize: arrival arrays -> CSR ----------------
 static dg_status finalize_csr(dg_index* ix) {
-  const int32_t nlist =
-      ix->desc.kind == DG_INDEX_FLAT ? 1 : ix->desc.nlist;
+  const bool is_ivf = ix->desc.kind == DG_INDEX_IVF_FLAT;
+  const int32_t nlist = is_ivf ? ix->desc.nlist : 1;
   const int32_t d = ix->desc.d;
   const int64_t n = ix->ntotal;
+  const int32_t CR = dg_index::kChunkRows;
   dg_status st;
+  // row-major target: persistent for FLAT (rocBLAS dots), temporary for IVF
+  dg_dbuf rm_tmp{};
+  dg_dbuf& rowmajor = is_ivf ? rm_tmp : ix->d_csr_vectors;
   if ((st = dbuf_reserve(ix->d_csr_offsets, ((size_t)nlist + 1) * 8,
                          ix->stream, false)) != DG_OK ||
-      (st = dbuf_reserve(ix->d_csr_vectors, (size_t)n * d * 4, ix->stream,
-                         false)) != DG_OK ||
+      (st = dbuf_reserve(rowmajor, (size_t)n * d * 4, ix->stream, false)) !=
+          DG_OK ||
       (st = dbuf_reserve(ix->d_csr_ids, (size_t)n * 8, ix->stream, false)) !=
           DG_OK ||
       (st = dbuf_reserve(ix->d_csr_vnorms, (size_t)n * 4, ix->stream,
                          false)) != DG_OK ||
       (st = dbuf_reserve(ix->ws_small,
                          (size_t)nlist * 4 + (size_t)n * 4 + 64, ix->stream,
-                         false)) != DG_OK)
+                         false)) != DG_OK) {
+    dbuf_free(rm_tmp);
     return st;
+  }
   int32_t* d_counts = (int32_t*)ix->ws_small.p;
   uint32_t* d_perm = (uint32_t*)((char*)ix->ws_small.p + (size_t)nlist * 4);
   int64_t* d_offsets = (int64_t*)ix->d_csr_offsets.p;
@@ -635,10 +641,10 @@ static dg_status finalize_csr(dg_index* ix) {
     dgk::scatter_perm(ix->stream, (const int32_t*)ix->d_assign.p, n, nullptr,
                       d_counts, d_perm);
     dgk::gather_rows(ix->stream, (const float*)ix->d_vectors.p, d_perm, n, d,
-                     (float*)ix->d_csr_vectors.p);
+                     (float*)rowmajor.p);
     dgk::gather_ids(ix->stream, (const int64_t*)ix->d_ids.p, d_perm, n,
                     (int64_t*)ix->d_csr_ids.p);
-    dgk::row_norms(ix->stream, (const float*)ix->d_csr_vectors.p, n, d,
+    dgk::row_norms(ix->stream, (const float*)rowmajor.p, n, d,
                    (float*)ix->d_csr_vnorms.p);
   } else {
     (void)hipMemsetAsync(d_offsets, 0, ((size_t)nlist + 1) * 8, ix->stream);
@@ -648,6 +654,59 @@ static dg_status finalize_csr(dg_index* ix) {
                               ((size_t)nlist + 1) * 8, hipMemcpyDeviceToHost,
                               ix->stream));
   DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+
+  if (is_ivf && n > 0) {
+    // chunk geometry (host) + transpose into [d][nrows_pad] chunk layout
+    std::vector<int32_t> chunk_off(nlist + 1, 0);
+    std::vector<int64_t> chunk_base;
+    std::vector<uint32_t> all_units;
+    int64_t t_elems = 0;
+    for (int32_t l = 0; l < nlist; l++) {
+      int64_t len = ix->h_csr_offsets[l + 1] - ix->h_csr_offsets[l];
+      int32_t nch = (int32_t)((len + CR - 1) / CR);
+      chunk_off[l + 1] = chunk_off[l] + nch;
+      for (int32_t c = 0; c < nch; c++) {
+        int32_t nrows = (int32_t)std::min<int64_t>(CR, len - (int64_t)c * CR);
+        int32_t pad = (nrows + 3) & ~3;
+        chunk_base.push_back(t_elems);
+        all_units.push_back((uint32_t)l);
+        all_units.push_back((uint32_t)c);
+        t_elems += (int64_t)d * pad;
+      }
+    }
+    ix->total_chunks = chunk_off[nlist];
+    size_t meta_bytes = ((size_t)nlist + 1) * 4 +
+                        (size_t)ix->total_chunks * 8 +
+                        (size_t)ix->total_chunks * 8;
+    if ((st = dbuf_reserve(ix->d_chunk_meta, meta_bytes, ix->stream,
+                           false)) != DG_OK ||
+        (st = dbuf_reserve(ix->d_csr_t, (size_t)t_elems * 4, ix->stream,
+                           false)) != DG_OK) {
+      dbuf_free(rm_tmp);
+      return st;
+    }
+    char* mp = (char*)ix->d_chunk_meta.p;
+    int32_t* d_chunk_off = (int32_t*)mp;
+    int64_t* d_chunk_base = (int64_t*)(mp + ((size_t)nlist + 1) * 4);
+    uint32_t* d_all_units =
+        (uint32_t*)(mp + ((size_t)nlist + 1) * 4 +
+                    (size_t)ix->total_chunks * 8);
+    (void)hipMemcpyAsync(d_chunk_off, chunk_off.data(),
+                         ((size_t)nlist + 1) * 4, hipMemcpyHostToDevice,
+                         ix->stream);
+    (void)hipMemcpyAsync(d_chunk_base, chunk_base.data(),
+                         (size_t)ix->total_chunks * 8, hipMemcpyHostToDevice,
+                         ix->stream);
+    (void)hipMemcpyAsync(d_all_units, all_units.data(),
+                         (size_t)ix->total_chunks * 8, hipMemcpyHostToDevice,
+                         ix->stream);
+    dgk::transpose_chunks(ix->stream, d_all_units, ix->total_chunks,
+                          d_offsets, d_chunk_off, d_chunk_base,
+                          (const float*)rowmajor.p, d, CR,
+                          (float*)ix->d_csr_t.p);
+    DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  }
+  dbuf_free(rm_tmp);
   ix->csr_valid = true;
   return DG_OK;
 }
@@ -851,7 +910,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                       (const int64_t*)ix->d_csr_offsets.p, qp_off, q_total);
     dgk::excl_scan_i64(ix->stream, q_total, nq, q_cand_base);
     // units
-    const int32_t chunk_rows = 1024;
+    const int32_t chunk_rows = dg_index::kChunkRows;
     dgk::fill_unit_counts(ix->stream, inv_counts, nlist,
                           (const int64_t*)ix->d_csr_offsets.p, chunk_rows,
                           unit_counts);
@@ -903,16 +962,18 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     dgk::fill_units(ix->stream, cursors, unit_counts, nlist, nullptr, 0,
                     units, total_units);
 
-    // THE scan
-    int32_t qt_max = std::max(1, std::min(16, (int32_t)(131072 / (4 * d))));
+    // THE scan (columnar v2)
+    char* mp = (char*)ix->d_chunk_meta.p;
+    const int32_t* d_chunk_off = (const int32_t*)mp;
+    const int64_t* d_chunk_base =
+        (const int64_t*)(mp + ((size_t)nlist + 1) * 4);
     (void)hipEventRecord(ix->ev[2], ix->stream);
-    dgk::ivf_scan(ix->stream, units, total_units,
-                  (const int64_t*)ix->d_csr_offsets.p,
-                  (const float*)ix->d_csr_vectors.p,
-                  (const float*)ix->d_csr_vnorms.p, nullptr, dq, nullptr, d,
-                  inv_offsets32, inv_q, inv_rank, qp_off, q_cand_base, metric,
-                  d_bitmap, chunk_rows, np, qt_max,
-                  (uint64_t*)ix->ws_cand.p);
+    dgk::ivf_scan_col(ix->stream, units, total_units,
+                      (const int64_t*)ix->d_csr_offsets.p, d_chunk_off,
+                      d_chunk_base, (const float*)ix->d_csr_t.p,
+                      (const float*)ix->d_csr_vnorms.p, dq, d, inv_offsets32,
+                      inv_q, inv_rank, qp_off, q_cand_base, np, metric,
+                      d_bitmap, chunk_rows, (uint64_t*)ix->ws_cand.p);
     (void)hipEventRecord(ix->ev[3], ix->stream);
 
     // select + emit

@@ -85,10 +85,17 @@ struct dg_index {
   // finalized CSR (built lazily from arrival store on first search)
   bool csr_valid = false;
   dg_dbuf d_csr_offsets;    // [nlist+1] i64
-  dg_dbuf d_csr_vectors;    // [ntotal x d] grouped by list
+  dg_dbuf d_csr_vectors;    // FLAT: [ntotal x d] row-major (rocBLAS dots)
   dg_dbuf d_csr_ids;        // [ntotal] i64 (-2 = deleted tombstone)
-  dg_dbuf d_csr_vnorms;     // [ntotal] f32 norms of grouped vectors
+  dg_dbuf d_csr_vnorms;     // [ntotal] f32 norms in CSR row order
   std::vector<int64_t> h_csr_offsets;  // host copy for planning
+  // IVF: column-major 1024-row chunks ([d][nrows_pad] per chunk) — the
+  // layout the v2 scan kernel reads
+  dg_dbuf d_csr_t;          // transposed chunk data
+  dg_dbuf d_chunk_meta;     // chunk_off i32[nlist+1] + chunk_base i64[nchunks]
+                            // + all-chunks unit array u32[2*nchunks]
+  int32_t total_chunks = 0;
+  static constexpr int32_t kChunkRows = 1024;
 
   // Flat uses d_csr_* with a single implicit list (nlist=1) so scan/select
   // machinery is shared.
@@ -170,6 +177,18 @@ void ivf_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
               const int64_t* q_cand_base, int metric,
               const uint32_t* pass_bitmap, int32_t chunk_rows, int32_t nprobe,
               int32_t qt_max, uint64_t* cand);
+void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
+                  const int64_t* csr_offsets, const int32_t* chunk_off,
+                  const int64_t* chunk_base, const float* tvec,
+                  const float* vnorms, const float* queries, int32_t d,
+                  const int32_t* inv_offsets, const int32_t* inv_q,
+                  const int32_t* inv_rank, const int64_t* qp_off,
+                  const int64_t* q_cand_base, int32_t nprobe, int metric,
+                  const uint32_t* bitmap, int32_t chunk_rows, uint64_t* cand);
+void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
+                      const int64_t* csr_offsets, const int32_t* chunk_off,
+                      const int64_t* chunk_base, const float* rowmajor,
+                      int32_t d, int32_t chunk_rows, float* tvec);
 // emit: resolve ids, apply metric convention
 void emit_results(hipStream_t s, const uint64_t* topk,
                   const int64_t* ids_lookup, const float* qnorms, int64_t nq,

@@ -494,6 +494,162 @@ __global__ void k_tombstone(const int64_t* __restrict__ pos, int64_t n,
   if (i < n) ids[pos[i]] = -2;
 }
 
+// ---------- columnar scan (v2) ----------
+// The dominant kernel, redesigned: CSR chunks are stored TRANSPOSED
+// ([d][nrows_pad] per 1024-row chunk).  Each lane owns
+// 4 consecutive rows and accumulates full dots for QTM LDS-staged queries
+// in registers — no cross-lane reduction at all (the v1 wave-per-row form
+// was bound by its 6-deep dependent shuffle chain per (row, query)).
+// Column loads are 16 B/lane coalesced; query element broadcasts come from
+// LDS.  Unused query slots are zero-staged so the inner loop is branch-free.
+template <int QTM, int RPL>  // queries per tile, rows per lane
+__global__ void __launch_bounds__(256, 2) k_ivf_scan_col(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  extern __shared__ float smem[];          // [QTM * d] query tile
+  int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);  // [QTM]
+
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* col = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+  const int64_t row0 = list_start + (int64_t)chunk * chunk_rows;
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+
+  for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
+    const int32_t qt = min(QTM, nql - t0);
+    __syncthreads();
+    for (int32_t j = 0; j < qt; j++) {
+      int32_t q = inv_q[iq0 + t0 + j];
+      const float4* src = (const float4*)(queries + (size_t)q * d);
+      float4* dst = (float4*)(smem + (size_t)j * d);
+      for (int i = threadIdx.x; i < d / 4; i += blockDim.x) dst[i] = src[i];
+    }
+    // zero-stage unused slots (keeps the inner loop branch-free)
+    for (size_t i = (size_t)qt * d + threadIdx.x; i < (size_t)QTM * d;
+         i += blockDim.x)
+      smem[i] = 0.f;
+    if (threadIdx.x < QTM) {
+      int32_t j = threadIdx.x;
+      if (j < qt) {
+        int32_t q = inv_q[iq0 + t0 + j];
+        int32_t rank = inv_rank[iq0 + t0 + j];
+        cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                   list_start;
+      } else {
+        cbase[j] = 0;
+      }
+    }
+    __syncthreads();
+
+    // each wave covers WAVE*RPL consecutive rows; waves stride the chunk
+    for (int32_t rb = wave_id * WAVE * RPL; rb < nrows_pad;
+         rb += 4 * WAVE * RPL) {
+      const int32_t rr0 = rb + lane * RPL;  // this lane's RPL rows
+      if (rr0 >= nrows_pad) continue;
+      float acc[QTM][RPL];
+#pragma unroll
+      for (int j = 0; j < QTM; j++)
+#pragma unroll
+        for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
+
+      for (int32_t i = 0; i < d; i++) {
+        const float* cp = col + (size_t)i * nrows_pad + rr0;
+        float c[RPL];
+        if (RPL == 4) {
+          const float4 c4 = *(const float4*)cp;
+          c[0] = c4.x; c[1] = c4.y; c[2] = c4.z; c[RPL - 1] = c4.w;
+        } else if (RPL == 2) {
+          const float2 c2 = *(const float2*)cp;
+          c[0] = c2.x; c[RPL - 1] = c2.y;
+        } else {
+          c[0] = *cp;
+        }
+#pragma unroll
+        for (int j = 0; j < QTM; j++) {
+          const float qv = smem[(size_t)j * d + i];
+#pragma unroll
+          for (int x = 0; x < RPL; x++) acc[j][x] += c[x] * qv;
+        }
+      }
+
+      // emit: lane's rows are consecutive in each query's segment
+#pragma unroll
+      for (int j = 0; j < QTM; j++) {
+        if (j >= qt) break;
+        const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+        for (int x = 0; x < RPL; x++) {
+          const int32_t rl = rr0 + x;
+          if (rl >= nrows) break;
+          const int64_t r = row0 + rl;
+          bool pass = true;
+          if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+          float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                    : -acc[j][x];
+          cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+        }
+      }
+    }
+  }
+}
+
+// tiled row-major -> column-major chunk transpose (finalize step).
+// One block per (list, chunk) unit over ALL chunks; LDS 64x65 tile.
+__global__ void k_transpose_chunks(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ rowmajor, int32_t d, int32_t chunk_rows,
+    float* __restrict__ tvec) {
+  __shared__ float tile[64][65];
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* src =
+      rowmajor + (list_start + (int64_t)chunk * chunk_rows) * d;
+  float* dst = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+
+  const int tx = threadIdx.x % 64;  // dim within tile on read, row on write
+  const int ty = threadIdx.x / 64;  // 4 groups
+  for (int32_t r0 = 0; r0 < nrows; r0 += 64) {
+    for (int32_t i0 = 0; i0 < d; i0 += 64) {
+      __syncthreads();
+      for (int rr = ty; rr < 64; rr += 4) {
+        int32_t r = r0 + rr;
+        tile[rr][tx] = (r < nrows && i0 + tx < d)
+                           ? src[(size_t)r * d + i0 + tx]
+                           : 0.f;
+      }
+      __syncthreads();
+      for (int ii = ty; ii < 64; ii += 4) {
+        int32_t i = i0 + ii;
+        int32_t r = r0 + tx;
+        if (i < d && r < nrows_pad)
+          dst[(size_t)i * nrows_pad + r] = tile[tx][ii];
+      }
+    }
+  }
+}
+
 // ---------- k-means / finalize helpers ----------
 __global__ void k_hist_assign(const int32_t* __restrict__ assign, int64_t n,
                               int32_t* __restrict__ counts) {
@@ -769,6 +925,49 @@ void ivf_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
       break;  // host validates d <= 2048 for IVF scan
   }
 #undef DG_SCAN_CASE
+}
+
+void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
+                  const int64_t* csr_offsets, const int32_t* chunk_off,
+                  const int64_t* chunk_base, const float* tvec,
+                  const float* vnorms, const float* queries, int32_t d,
+                  const int32_t* inv_offsets, const int32_t* inv_q,
+                  const int32_t* inv_rank, const int64_t* qp_off,
+                  const int64_t* q_cand_base, int32_t nprobe, int metric,
+                  const uint32_t* bitmap, int32_t chunk_rows,
+                  uint64_t* cand) {
+  if (!n_units) return;
+  static int variant = []() {
+    const char* e = getenv("DG_SCAN_VARIANT");
+    return e ? atoi(e) : 0;
+  }();
+  if (variant == 1) {
+    constexpr int QTM = 8;
+    size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+    hipLaunchKernelGGL((k_ivf_scan_col<QTM, 4>), dim3((uint32_t)n_units),
+                       dim3(256), lds, s, units, csr_offsets, chunk_off,
+                       chunk_base, tvec, vnorms, queries, d, inv_offsets,
+                       inv_q, inv_rank, qp_off, q_cand_base, nprobe, metric,
+                       bitmap, chunk_rows, cand);
+  } else {
+    constexpr int QTM = 16;
+    size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+    hipLaunchKernelGGL((k_ivf_scan_col<QTM, 2>), dim3((uint32_t)n_units),
+                       dim3(256), lds, s, units, csr_offsets, chunk_off,
+                       chunk_base, tvec, vnorms, queries, d, inv_offsets,
+                       inv_q, inv_rank, qp_off, q_cand_base, nprobe, metric,
+                       bitmap, chunk_rows, cand);
+  }
+}
+
+void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
+                      const int64_t* csr_offsets, const int32_t* chunk_off,
+                      const int64_t* chunk_base, const float* rowmajor,
+                      int32_t d, int32_t chunk_rows, float* tvec) {
+  if (!n_units) return;
+  hipLaunchKernelGGL(k_transpose_chunks, dim3((uint32_t)n_units), dim3(256),
+                     0, s, units, csr_offsets, chunk_off, chunk_base,
+                     rowmajor, d, chunk_rows, tvec);
 }
 
 void hist_assign(hipStream_t s, const int32_t* assign, int64_t n,
