@@ -503,7 +503,7 @@ __global__ void k_tombstone(const int64_t* __restrict__ pos, int64_t n,
 // Column loads are 16 B/lane coalesced; query element broadcasts come from
 // LDS.  Unused query slots are zero-staged so the inner loop is branch-free.
 template <int QTM, int RPL>  // queries per tile, rows per lane
-__global__ void __launch_bounds__(256, 2) k_ivf_scan_col(
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
     const float* __restrict__ tvec, const float* __restrict__ vnorms,
@@ -571,9 +571,15 @@ __global__ void __launch_bounds__(256, 2) k_ivf_scan_col(
       for (int32_t i = 0; i < d; i++) {
         const float* cp = col + (size_t)i * nrows_pad + rr0;
         float c[RPL];
-        if (RPL == 4) {
-          const float4 c4 = *(const float4*)cp;
-          c[0] = c4.x; c[1] = c4.y; c[2] = c4.z; c[RPL - 1] = c4.w;
+        if (RPL >= 4) {
+#pragma unroll
+          for (int v = 0; v < RPL / 4; v++) {
+            const float4 c4 = ((const float4*)cp)[v];
+            c[4 * v + 0] = c4.x;
+            c[4 * v + 1] = c4.y;
+            c[4 * v + 2] = c4.z;
+            c[4 * v + 3] = c4.w;
+          }
         } else if (RPL == 2) {
           const float2 c2 = *(const float2*)cp;
           c[0] = c2.x; c[RPL - 1] = c2.y;
@@ -589,20 +595,24 @@ __global__ void __launch_bounds__(256, 2) k_ivf_scan_col(
       }
 
       // emit: lane's rows are consecutive in each query's segment
+      // (guards, not breaks: a break blocks full unroll and acc[] would be
+      // dynamically indexed -> scratch spill)
 #pragma unroll
       for (int j = 0; j < QTM; j++) {
-        if (j >= qt) break;
-        const int64_t cb = cbase[j] + row0 + rr0;
+        if (j < qt) {
+          const int64_t cb = cbase[j] + row0 + rr0;
 #pragma unroll
-        for (int x = 0; x < RPL; x++) {
-          const int32_t rl = rr0 + x;
-          if (rl >= nrows) break;
-          const int64_t r = row0 + rl;
-          bool pass = true;
-          if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
-          float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
-                                    : -acc[j][x];
-          cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+          for (int x = 0; x < RPL; x++) {
+            const int32_t rl = rr0 + x;
+            if (rl < nrows) {
+              const int64_t r = row0 + rl;
+              bool pass = true;
+              if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+              float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                        : -acc[j][x];
+              cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+            }
+          }
         }
       }
     }
@@ -941,23 +951,24 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     const char* e = getenv("DG_SCAN_VARIANT");
     return e ? atoi(e) : 0;
   }();
-  if (variant == 1) {
-    constexpr int QTM = 8;
-    size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-    hipLaunchKernelGGL((k_ivf_scan_col<QTM, 4>), dim3((uint32_t)n_units),
-                       dim3(256), lds, s, units, csr_offsets, chunk_off,
-                       chunk_base, tvec, vnorms, queries, d, inv_offsets,
-                       inv_q, inv_rank, qp_off, q_cand_base, nprobe, metric,
-                       bitmap, chunk_rows, cand);
-  } else {
-    constexpr int QTM = 16;
-    size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-    hipLaunchKernelGGL((k_ivf_scan_col<QTM, 2>), dim3((uint32_t)n_units),
-                       dim3(256), lds, s, units, csr_offsets, chunk_off,
-                       chunk_base, tvec, vnorms, queries, d, inv_offsets,
-                       inv_q, inv_rank, qp_off, q_cand_base, nprobe, metric,
-                       bitmap, chunk_rows, cand);
+#define DG_SCAN_LAUNCH(QTM, RPL)                                            \
+  do {                                                                      \
+    size_t lds = (size_t)(QTM) * d * 4 + (QTM) * 8;                         \
+    hipLaunchKernelGGL((k_ivf_scan_col<(QTM), (RPL)>),                      \
+                       dim3((uint32_t)n_units), dim3(256), lds, s, units,   \
+                       csr_offsets, chunk_off, chunk_base, tvec, vnorms,    \
+                       queries, d, inv_offsets, inv_q, inv_rank, qp_off,    \
+                       q_cand_base, nprobe, metric, bitmap, chunk_rows,     \
+                       cand);                                               \
+  } while (0)
+  switch (variant) {
+    case 1: DG_SCAN_LAUNCH(8, 4); break;
+    case 2: DG_SCAN_LAUNCH(16, 2); break;
+    case 3: DG_SCAN_LAUNCH(8, 8); break;
+    case 4: DG_SCAN_LAUNCH(4, 4); break;
+    default: DG_SCAN_LAUNCH(16, 4); break;
   }
+#undef DG_SCAN_LAUNCH
 }
 
 void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
