@@ -568,29 +568,55 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
 #pragma unroll
         for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
 
-      for (int32_t i = 0; i < d; i++) {
-        const float* cp = col + (size_t)i * nrows_pad + rr0;
-        float c[RPL];
-        if (RPL >= 4) {
+      // dim loop unrolled by U with the U column loads issued together
+      // (one load in flight per wave exposes full HBM latency per iter);
+      // query elements read 4-at-a-time from LDS (b128 halves LDS cycles)
+      constexpr int U = 8;
+      static_assert(RPL == 1 || RPL == 2 || RPL % 4 == 0, "RPL");
+      int32_t i = 0;
+      for (; i + U <= d; i += U) {
+        float c[U][RPL];
 #pragma unroll
-          for (int v = 0; v < RPL / 4; v++) {
-            const float4 c4 = ((const float4*)cp)[v];
-            c[4 * v + 0] = c4.x;
-            c[4 * v + 1] = c4.y;
-            c[4 * v + 2] = c4.z;
-            c[4 * v + 3] = c4.w;
+        for (int u = 0; u < U; u++) {
+          const float* cp = col + (size_t)(i + u) * nrows_pad + rr0;
+          if (RPL >= 4) {
+#pragma unroll
+            for (int v = 0; v < RPL / 4; v++) {
+              const float4 c4 = ((const float4*)cp)[v];
+              c[u][4 * v + 0] = c4.x;
+              c[u][4 * v + 1] = c4.y;
+              c[u][4 * v + 2] = c4.z;
+              c[u][4 * v + 3] = c4.w;
+            }
+          } else if (RPL == 2) {
+            const float2 c2 = *(const float2*)cp;
+            c[u][0] = c2.x; c[u][RPL - 1] = c2.y;
+          } else {
+            c[u][0] = *cp;
           }
-        } else if (RPL == 2) {
-          const float2 c2 = *(const float2*)cp;
-          c[0] = c2.x; c[RPL - 1] = c2.y;
-        } else {
-          c[0] = *cp;
         }
+#pragma unroll
+        for (int u4 = 0; u4 < U / 4; u4++) {
+#pragma unroll
+          for (int j = 0; j < QTM; j++) {
+            const float4 qv4 =
+                *(const float4*)(smem + (size_t)j * d + i + u4 * 4);
+            const float qv[4] = {qv4.x, qv4.y, qv4.z, qv4.w};
+#pragma unroll
+            for (int uu = 0; uu < 4; uu++)
+#pragma unroll
+              for (int x = 0; x < RPL; x++)
+                acc[j][x] += c[u4 * 4 + uu][x] * qv[uu];
+          }
+        }
+      }
+      for (; i < d; i++) {  // tail (d % 8, always a multiple of 4)
+        const float* cp = col + (size_t)i * nrows_pad + rr0;
 #pragma unroll
         for (int j = 0; j < QTM; j++) {
           const float qv = smem[(size_t)j * d + i];
 #pragma unroll
-          for (int x = 0; x < RPL; x++) acc[j][x] += c[x] * qv;
+          for (int x = 0; x < RPL; x++) acc[j][x] += cp[x] * qv;
         }
       }
 
