@@ -513,7 +513,7 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
     const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
     const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
     uint64_t* __restrict__ cand) {
-  extern __shared__ float smem[];          // [QTM * d] query tile
+  extern __shared__ __attribute__((aligned(16))) float smem[];  // [QTM * d]
   int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);  // [QTM]
 
   const uint32_t list = units[2 * blockIdx.x];
@@ -530,6 +530,7 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
 
   const int wave_id = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
+  __builtin_assume(d % 4 == 0 && d > 0);  // host enforces; enables b128 LDS
 
   for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
     const int32_t qt = min(QTM, nql - t0);
@@ -568,17 +569,19 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
 #pragma unroll
         for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
 
-      // dim loop unrolled by U with the U column loads issued together
-      // (one load in flight per wave exposes full HBM latency per iter);
-      // query elements read 4-at-a-time from LDS (b128 halves LDS cycles)
-      constexpr int U = 8;
+      // dim loop unrolled by U, software-pipelined: while block i computes,
+      // block i+U's loads are in flight (double-buffered column registers —
+      // waiting within the issuing iteration exposes full HBM latency).
+      // Query elements read 4-at-a-time from aligned LDS.
+      constexpr int U = 4;
       static_assert(RPL == 1 || RPL == 2 || RPL % 4 == 0, "RPL");
-      int32_t i = 0;
-      for (; i + U <= d; i += U) {
-        float c[U][RPL];
+      float ca[U][RPL], cb[U][RPL];
+      const float* colp = col + rr0;
+      const int32_t nfull = d / U;  // d % 4 == 0, U == 4 -> no tail
+      auto load_block = [&](float (&c)[U][RPL], int32_t ib) {
 #pragma unroll
         for (int u = 0; u < U; u++) {
-          const float* cp = col + (size_t)(i + u) * nrows_pad + rr0;
+          const float* cp = colp + (size_t)(ib + u) * nrows_pad;
           if (RPL >= 4) {
 #pragma unroll
             for (int v = 0; v < RPL / 4; v++) {
@@ -595,30 +598,27 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
             c[u][0] = *cp;
           }
         }
-#pragma unroll
-        for (int u4 = 0; u4 < U / 4; u4++) {
-#pragma unroll
-          for (int j = 0; j < QTM; j++) {
-            const float4 qv4 =
-                *(const float4*)(smem + (size_t)j * d + i + u4 * 4);
-            const float qv[4] = {qv4.x, qv4.y, qv4.z, qv4.w};
-#pragma unroll
-            for (int uu = 0; uu < 4; uu++)
-#pragma unroll
-              for (int x = 0; x < RPL; x++)
-                acc[j][x] += c[u4 * 4 + uu][x] * qv[uu];
-          }
-        }
-      }
-      for (; i < d; i++) {  // tail (d % 8, always a multiple of 4)
-        const float* cp = col + (size_t)i * nrows_pad + rr0;
+      };
+      auto compute_block = [&](float (&c)[U][RPL], int32_t ib) {
 #pragma unroll
         for (int j = 0; j < QTM; j++) {
-          const float qv = smem[(size_t)j * d + i];
+          const float4 qv4 = *(const float4*)(smem + (size_t)j * d + ib);
+          const float qv[4] = {qv4.x, qv4.y, qv4.z, qv4.w};
 #pragma unroll
-          for (int x = 0; x < RPL; x++) acc[j][x] += cp[x] * qv;
+          for (int uu = 0; uu < U; uu++)
+#pragma unroll
+            for (int x = 0; x < RPL; x++) acc[j][x] += c[uu][x] * qv[uu];
         }
+      };
+      load_block(ca, 0);
+      int32_t ib = 0;
+      for (; ib + U < d; ib += 2 * U) {
+        load_block(cb, ib + U);
+        compute_block(ca, ib);
+        if (ib + 2 * U < d) load_block(ca, ib + 2 * U);
+        compute_block(cb, ib + U);
       }
+      if (ib < d) compute_block(ca, ib);
 
       // emit: lane's rows are consecutive in each query's segment
       // (guards, not breaks: a break blocks full unroll and acc[] would be
