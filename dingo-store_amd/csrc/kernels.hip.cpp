@@ -610,6 +610,7 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
             for (int x = 0; x < RPL; x++) acc[j][x] += c[uu][x] * qv[uu];
         }
       };
+      // 2-deep rotation (3-deep costs a wave of occupancy and regresses)
       load_block(ca, 0);
       int32_t ib = 0;
       for (; ib + U < d; ib += 2 * U) {
