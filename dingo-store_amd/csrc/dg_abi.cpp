@@ -153,14 +153,24 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
     return DG_EINVAL;
   }
   if (desc.kind == DG_INDEX_IVF_PQ) {
-    dg_set_error("IVF_PQ kernels land in a later round (SURVEY.md §7 item 6)");
-    return DG_ENOT_SUPPORT;
+    if (desc.pq_m <= 0) desc.pq_m = 64;      // kCreateIvfPqParamNsubvector
+    if (desc.pq_nbits <= 0) desc.pq_nbits = 8;  // kCreateIvfPqParamNbitsPerIdx
+    if (desc.pq_nbits != 8) {
+      dg_set_error("only nbits=8 supported (reference default, constant.h)");
+      return DG_ENOT_SUPPORT;
+    }
+    if (desc.d % desc.pq_m != 0 || desc.pq_m % 4 != 0) {
+      dg_set_error("need d %% m == 0 and m %% 4 == 0 (d=%d m=%d)", desc.d,
+                   desc.pq_m);
+      return DG_ENOT_SUPPORT;
+    }
   }
-  if (desc.kind != DG_INDEX_FLAT && desc.kind != DG_INDEX_IVF_FLAT) {
+  if (desc.kind != DG_INDEX_FLAT && desc.kind != DG_INDEX_IVF_FLAT &&
+      desc.kind != DG_INDEX_IVF_PQ) {
     dg_set_error("bad kind %d", desc.kind);
     return DG_EINVAL;
   }
-  if (desc.kind == DG_INDEX_IVF_FLAT && desc.nlist <= 0)
+  if (desc.kind != DG_INDEX_FLAT && desc.nlist <= 0)
     desc.nlist = 2048;  // kCreateIvfFlatParamNcentroids, constant.h:177
   if (desc.kind == DG_INDEX_IVF_FLAT && desc.d > 2048) {
     dg_set_error("IVF scan kernel supports d <= 2048 this round");
@@ -206,7 +216,9 @@ extern "C" void dg_index_destroy(dg_index* ix) {
   for (auto b :
        {&ix->d_vectors, &ix->d_ids, &ix->d_assign, &ix->d_centroids,
         &ix->d_cnorms, &ix->d_csr_offsets, &ix->d_csr_vectors, &ix->d_csr_ids,
-        &ix->d_csr_vnorms, &ix->d_csr_t, &ix->d_chunk_meta, &ix->d_list_mask, &ix->ws_queries, &ix->ws_qnorms,
+        &ix->d_csr_vnorms, &ix->d_csr_t, &ix->d_chunk_meta, &ix->d_list_mask,
+        &ix->d_codebooks, &ix->d_codes, &ix->d_csr_codes, &ix->d_S,
+        &ix->d_cb_norms, &ix->ws_T, &ix->ws_queries, &ix->ws_qnorms,
         &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
         &ix->ws_small, &ix->ws_topk})
     dbuf_free(*b);
@@ -221,7 +233,7 @@ extern "C" void dg_index_destroy(dg_index* ix) {
 extern "C" dg_status dg_set_centroids(dg_index* ix, int32_t nlist,
                                       const float* centroids) {
   if (!ix || !centroids) return DG_EINVAL;
-  if (ix->desc.kind != DG_INDEX_IVF_FLAT) {
+  if (ix->desc.kind == DG_INDEX_FLAT) {
     dg_set_error("set_centroids on non-IVF index");
     return DG_EINVAL;
   }
@@ -250,7 +262,7 @@ extern "C" dg_status dg_set_centroids(dg_index* ix, int32_t nlist,
 
 extern "C" dg_status dg_get_centroids(dg_index* ix, float* out) {
   if (!ix || !out) return DG_EINVAL;
-  if (!ix->trained || ix->desc.kind != DG_INDEX_IVF_FLAT) {
+  if (!ix->trained || ix->desc.kind == DG_INDEX_FLAT) {
     dg_set_error("not trained");
     return DG_ENOT_TRAINED;
   }
@@ -264,26 +276,218 @@ extern "C" dg_status dg_get_centroids(dg_index* ix, float* out) {
 
 // ---------------- train (GPU k-means; faiss Clustering semantics
 // restated — see oracle.c for the algorithm statement) ----------------
-static dg_status assign_rows(dg_index* ix, const float* d_x, int64_t n,
-                             int32_t* d_assign) {
-  // chunked: dots[chunk x nlist] = X * C^T ; argmin
-  const int32_t nlist = ix->desc.nlist;
-  const int32_t d = ix->desc.d;
+static dg_status assign_rows_gen(dg_index* ix, const float* d_x, int64_t n,
+                                 int32_t dim, const float* d_cents,
+                                 const float* d_cnorms, int32_t k, int metric,
+                                 int32_t* d_assign) {
+  // chunked: dots[chunk x k] = X * C^T ; argmin
   const int64_t chunk = std::max<int64_t>(
-      1, std::min<int64_t>(n, (int64_t)(512ull << 20) / ((size_t)nlist * 4)));
+      1, std::min<int64_t>(n, (int64_t)(512ull << 20) / ((size_t)k * 4)));
   dg_status st =
-      dbuf_reserve(ix->ws_dots, (size_t)chunk * nlist * 4, ix->stream, false);
+      dbuf_reserve(ix->ws_dots, (size_t)chunk * k * 4, ix->stream, false);
   if (st != DG_OK) return st;
   for (int64_t s0 = 0; s0 < n; s0 += chunk) {
     int64_t c = std::min(chunk, n - s0);
-    st = sgemm_dots(ix, d_x + (size_t)s0 * d, c,
-                    (const float*)ix->d_centroids.p, nlist, d,
+    st = sgemm_dots(ix, d_x + (size_t)s0 * dim, c, d_cents, k, dim,
                     (float*)ix->ws_dots.p);
     if (st != DG_OK) return st;
-    dgk::argmin_rows(ix->stream, (const float*)ix->ws_dots.p,
-                     (const float*)ix->d_cnorms.p, c, nlist, ix->desc.metric,
-                     d_assign + s0);
+    dgk::argmin_rows(ix->stream, (const float*)ix->ws_dots.p, d_cnorms, c, k,
+                     metric, d_assign + s0);
   }
+  return DG_OK;
+}
+
+static dg_status assign_rows(dg_index* ix, const float* d_x, int64_t n,
+                             int32_t* d_assign) {
+  return assign_rows_gen(ix, d_x, n, ix->desc.d,
+                         (const float*)ix->d_centroids.p,
+                         (const float*)ix->d_cnorms.p, ix->desc.nlist,
+                         ix->desc.metric, d_assign);
+}
+
+// k-means over DEVICE data (faiss Clustering restated: subsample to 256*k,
+// init from perm(seed+1), 25 iters, empty-cluster split — oracle.c states
+// the rules).  d_cents: k*dim device output.
+static dg_status kmeans_device(dg_index* ix, const float* d_data_in,
+                               int64_t n_in, int32_t dim, int32_t k,
+                               int metric, uint32_t seed, float* d_cents) {
+  const int32_t niter = 25;
+  const int64_t maxp = (int64_t)256 * k;
+  dg_status st = DG_OK;
+  dg_dbuf d_sub{}, d_grouped{}, d_asg{}, d_perm{}, d_off{}, d_cn{}, d_idx{};
+  const float* d_data = d_data_in;
+  int64_t nt = n_in;
+  do {
+    if (n_in > maxp) {  // device-side subsample gather by host perm indices
+      std::vector<int64_t> perm;
+      rand_perm(perm, n_in, seed);
+      perm.resize(maxp);
+      if ((st = dbuf_reserve(d_idx, (size_t)maxp * 8, ix->stream, false)) !=
+              DG_OK ||
+          (st = dbuf_reserve(d_sub, (size_t)maxp * dim * 4, ix->stream,
+                             false)) != DG_OK)
+        break;
+      (void)hipMemcpyAsync(d_idx.p, perm.data(), (size_t)maxp * 8,
+                           hipMemcpyHostToDevice, ix->stream);
+      dgk::gather_rows_by_index(ix->stream, d_data_in,
+                                (const int64_t*)d_idx.p, maxp, dim,
+                                (float*)d_sub.p);
+      d_data = (const float*)d_sub.p;
+      nt = maxp;
+    }
+    if ((st = dbuf_reserve(d_grouped, (size_t)nt * dim * 4, ix->stream,
+                           false)) != DG_OK ||
+        (st = dbuf_reserve(d_asg, (size_t)nt * 4, ix->stream, false)) !=
+            DG_OK ||
+        (st = dbuf_reserve(d_perm, (size_t)nt * 4, ix->stream, false)) !=
+            DG_OK ||
+        (st = dbuf_reserve(d_off, ((size_t)k + 1) * 8 + (size_t)k * 4 + 64,
+                           ix->stream, false)) != DG_OK ||
+        (st = dbuf_reserve(d_cn, (size_t)k * 4, ix->stream, false)) != DG_OK)
+      break;
+    int64_t* d_offsets = (int64_t*)d_off.p;
+    int32_t* d_counts = (int32_t*)(d_offsets + k + 1);
+
+    {  // init: first k rows of perm(seed+1)
+      std::vector<int64_t> p2;
+      rand_perm(p2, nt, seed + 1);
+      p2.resize(k);
+      if ((st = dbuf_reserve(d_idx, (size_t)k * 8, ix->stream, false)) !=
+          DG_OK)
+        break;
+      (void)hipMemcpyAsync(d_idx.p, p2.data(), (size_t)k * 8,
+                           hipMemcpyHostToDevice, ix->stream);
+      dgk::gather_rows_by_index(ix->stream, d_data, (const int64_t*)d_idx.p,
+                                k, dim, d_cents);
+    }
+
+    Mt19937 split_rng(seed + 2);
+    std::vector<float> h_cents((size_t)k * dim);
+    std::vector<int64_t> h_off(k + 1);
+    for (int32_t iter = 0; iter < niter && st == DG_OK; iter++) {
+      dgk::row_norms(ix->stream, d_cents, k, dim, (float*)d_cn.p);
+      st = assign_rows_gen(ix, d_data, nt, dim, d_cents,
+                           (const float*)d_cn.p, k, metric,
+                           (int32_t*)d_asg.p);
+      if (st != DG_OK) break;
+      (void)hipMemsetAsync(d_counts, 0, (size_t)k * 4, ix->stream);
+      dgk::hist_assign(ix->stream, (const int32_t*)d_asg.p, nt, k, d_counts);
+      dgk::excl_scan_i32_to_i64(ix->stream, d_counts, k, d_offsets);
+      dgk::init_cursors(ix->stream, d_offsets, k, d_counts);
+      dgk::scatter_perm(ix->stream, (const int32_t*)d_asg.p, nt, nullptr,
+                        d_counts, (uint32_t*)d_perm.p);
+      dgk::gather_rows(ix->stream, d_data, (const uint32_t*)d_perm.p, nt,
+                       dim, (float*)d_grouped.p);
+      dgk::cluster_means(ix->stream, (const float*)d_grouped.p, d_offsets, k,
+                         dim, d_cents);
+      (void)hipMemcpyAsync(h_off.data(), d_offsets, ((size_t)k + 1) * 8,
+                           hipMemcpyDeviceToHost, ix->stream);
+      if (hipStreamSynchronize(ix->stream) != hipSuccess) {
+        st = DG_EINTERNAL;
+        break;
+      }
+      std::vector<int64_t> hist(k);
+      bool any_empty = false;
+      for (int32_t l = 0; l < k; l++) {
+        hist[l] = h_off[l + 1] - h_off[l];
+        if (hist[l] == 0) any_empty = true;
+      }
+      if (any_empty) {
+        (void)hipMemcpy(h_cents.data(), d_cents, (size_t)k * dim * 4,
+                        hipMemcpyDeviceToHost);
+        const float EPS = 1.0f / 1024.0f;
+        for (int32_t ci = 0; ci < k; ci++) {
+          if (hist[ci] != 0) continue;
+          int32_t cj = 0;
+          for (;; cj = (cj + 1) % k) {
+            float p = (hist[cj] - 1.0f) / (float)(nt - k);
+            if (split_rng.rand_float() < p) break;
+          }
+          memcpy(&h_cents[(size_t)ci * dim], &h_cents[(size_t)cj * dim],
+                 (size_t)dim * 4);
+          for (int32_t j = 0; j < dim; j++) {
+            float sgn = (j % 2 == 0) ? 1 + EPS : 1 - EPS;
+            h_cents[(size_t)ci * dim + j] *= sgn;
+            h_cents[(size_t)cj * dim + j] *= 2 - sgn;
+          }
+          hist[ci] = hist[cj] / 2;
+          hist[cj] -= hist[ci];
+        }
+        (void)hipMemcpy(d_cents, h_cents.data(), (size_t)k * dim * 4,
+                        hipMemcpyHostToDevice);
+      }
+    }
+  } while (0);
+  dbuf_free(d_sub);
+  dbuf_free(d_grouped);
+  dbuf_free(d_asg);
+  dbuf_free(d_perm);
+  dbuf_free(d_off);
+  dbuf_free(d_cn);
+  dbuf_free(d_idx);
+  return st;
+}
+
+// build the index-static PQ tables: S = ||c_m + cb||^2 and codebook-entry
+// norms (encode argmin keys)
+static dg_status finish_pq_tables(dg_index* ix) {
+  const int32_t M = ix->desc.pq_m;
+  const int32_t d = ix->desc.d;
+  const int32_t dsub = d / M;
+  const int32_t nlist = ix->desc.nlist;
+  dg_status st;
+  if ((st = dbuf_reserve(ix->d_S, (size_t)nlist * M * 256 * 4, ix->stream,
+                         false)) != DG_OK ||
+      (st = dbuf_reserve(ix->d_cb_norms, (size_t)M * 256 * 4, ix->stream,
+                         false)) != DG_OK)
+    return st;
+  dgk::build_S(ix->stream, (const float*)ix->d_centroids.p,
+               (const float*)ix->d_codebooks.p, nlist, M, dsub, d,
+               (float*)ix->d_S.p);
+  dgk::row_norms(ix->stream, (const float*)ix->d_codebooks.p,
+                 (int64_t)M * 256, dsub, (float*)ix->d_cb_norms.p);
+  return DG_OK;
+}
+
+extern "C" dg_status dg_set_codebooks(dg_index* ix, int32_t m, int32_t nbits,
+                                      const float* codebooks) {
+  if (!ix || !codebooks) return DG_EINVAL;
+  if (ix->desc.kind != DG_INDEX_IVF_PQ || m != ix->desc.pq_m || nbits != 8) {
+    dg_set_error("codebook shape mismatch (m=%d nbits=%d)", m, nbits);
+    return DG_EINVAL;
+  }
+  std::unique_lock lk(ix->rw);
+  if (!ix->trained) {
+    dg_set_error("set centroids before codebooks");
+    return DG_ENOT_TRAINED;
+  }
+  DeviceGuard g(ix->device);
+  const int32_t dsub = ix->desc.d / m;
+  size_t bytes = (size_t)m * 256 * dsub * 4;
+  dg_status st = dbuf_reserve(ix->d_codebooks, bytes, ix->stream, false);
+  if (st != DG_OK) return st;
+  DG_HIP_CHECK(hipMemcpyAsync(ix->d_codebooks.p, codebooks, bytes,
+                              hipMemcpyHostToDevice, ix->stream));
+  st = finish_pq_tables(ix);
+  if (st != DG_OK) return st;
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  ix->pq_trained = true;
+  ix->csr_valid = false;
+  return DG_OK;
+}
+
+extern "C" dg_status dg_get_codebooks(dg_index* ix, float* out) {
+  if (!ix || !out) return DG_EINVAL;
+  if (ix->desc.kind != DG_INDEX_IVF_PQ || !ix->pq_trained) {
+    dg_set_error("PQ not trained");
+    return DG_ENOT_TRAINED;
+  }
+  std::shared_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  const int32_t dsub = ix->desc.d / ix->desc.pq_m;
+  DG_HIP_CHECK(hipMemcpy(out, ix->d_codebooks.p,
+                         (size_t)ix->desc.pq_m * 256 * dsub * 4,
+                         hipMemcpyDeviceToHost));
   return DG_OK;
 }
 
@@ -301,147 +505,138 @@ extern "C" dg_status dg_train(dg_index* ix, int64_t n, const float* x) {
   if (n < ix->desc.nlist) ix->desc.nlist = 1;
   const int32_t nlist = ix->desc.nlist;
   const uint32_t seed = 1234;  // faiss ClusteringParameters.seed
-  const int32_t niter = 25;
 
-  // subsample to 256*nlist on host (faiss max_points_per_centroid)
-  const int64_t maxp = (int64_t)256 * nlist;
-  std::vector<int64_t> sel;
-  int64_t nt = n;
-  const float* src = x;
-  std::vector<float> staged;
-  if (n > maxp) {
-    std::vector<int64_t> perm;
-    rand_perm(perm, n, seed);
-    nt = maxp;
-    staged.resize((size_t)nt * d);
-    for (int64_t i = 0; i < nt; i++)
-      memcpy(&staged[(size_t)i * d], x + (size_t)perm[i] * d, (size_t)d * 4);
-    src = staged.data();
-  }
-  // upload train set
-  dg_dbuf d_td{}, d_grouped{}, d_asg{}, d_perm{}, d_off{};
-  dg_status st = dbuf_reserve(d_td, (size_t)nt * d * 4, ix->stream, false);
-  if (st == DG_OK)
-    st = dbuf_reserve(d_grouped, (size_t)nt * d * 4, ix->stream, false);
-  if (st == DG_OK) st = dbuf_reserve(d_asg, (size_t)nt * 4, ix->stream, false);
-  if (st == DG_OK) st = dbuf_reserve(d_perm, (size_t)nt * 4, ix->stream, false);
-  if (st == DG_OK)
-    st = dbuf_reserve(d_off, ((size_t)nlist + 2) * 8 + (size_t)nlist * 8,
-                      ix->stream, false);
-  if (st == DG_OK)
-    st = dbuf_reserve(ix->d_centroids, (size_t)nlist * d * 4, ix->stream,
-                      false);
-  if (st == DG_OK)
-    st = dbuf_reserve(ix->d_cnorms, (size_t)nlist * 4, ix->stream, false);
-  if (st != DG_OK) {
-    dbuf_free(d_td);
-    dbuf_free(d_grouped);
-    dbuf_free(d_asg);
-    dbuf_free(d_perm);
-    dbuf_free(d_off);
-    return st;
-  }
-  int64_t* d_offsets = (int64_t*)d_off.p;              // nlist+1
-  int32_t* d_counts = (int32_t*)(d_offsets + nlist + 1);  // nlist (reuse tail)
+  // upload train data to device (chunked)
+  dg_dbuf d_td{};
+  dg_status st = dbuf_reserve(d_td, (size_t)n * d * 4, ix->stream, false);
+  if (st != DG_OK) return st;
   do {
-    if (hipMemcpyAsync(d_td.p, src, (size_t)nt * d * 4, hipMemcpyHostToDevice,
+    if (hipMemcpyAsync(d_td.p, x, (size_t)n * d * 4, hipMemcpyHostToDevice,
                        ix->stream) != hipSuccess) {
       st = DG_EINTERNAL;
       break;
     }
     if (ix->desc.metric == DG_METRIC_COSINE)
-      dgk::normalize_rows(ix->stream, (float*)d_td.p, nt, d);
-
-    // init centroids: first nlist of perm(seed+1) of the (subsampled) set
-    {
-      std::vector<int64_t> p2;
-      rand_perm(p2, nt, seed + 1);
-      p2.resize(nlist);
-      dg_dbuf d_idx{};
-      st = dbuf_reserve(d_idx, (size_t)nlist * 8, ix->stream, false);
-      if (st != DG_OK) break;
-      (void)hipMemcpyAsync(d_idx.p, p2.data(), (size_t)nlist * 8,
-                           hipMemcpyHostToDevice, ix->stream);
-      dgk::gather_rows_by_index(ix->stream, (const float*)d_td.p,
-                                (const int64_t*)d_idx.p, nlist, d,
-                                (float*)ix->d_centroids.p);
-      (void)hipStreamSynchronize(ix->stream);
-      dbuf_free(d_idx);
-    }
-
-    Mt19937 split_rng(seed + 2);
-    std::vector<int32_t> h_counts(nlist);
-    std::vector<float> h_cents((size_t)nlist * d);
-    for (int32_t iter = 0; iter < niter && st == DG_OK; iter++) {
-      dgk::row_norms(ix->stream, (const float*)ix->d_centroids.p, nlist, d,
-                     (float*)ix->d_cnorms.p);
-      st = assign_rows(ix, (const float*)d_td.p, nt, (int32_t*)d_asg.p);
-      if (st != DG_OK) break;
-      (void)hipMemsetAsync(d_counts, 0, (size_t)nlist * 4, ix->stream);
-      dgk::hist_assign(ix->stream, (const int32_t*)d_asg.p, nt, nlist,
-                       d_counts);
-      dgk::excl_scan_i32_to_i64(ix->stream, d_counts, nlist, d_offsets);
-      dgk::init_cursors(ix->stream, d_offsets, nlist, d_counts);  // reuse
-      dgk::scatter_perm(ix->stream, (const int32_t*)d_asg.p, nt, nullptr,
-                        d_counts, (uint32_t*)d_perm.p);
-      dgk::gather_rows(ix->stream, (const float*)d_td.p,
-                       (const uint32_t*)d_perm.p, nt, d, (float*)d_grouped.p);
-      dgk::cluster_means(ix->stream, (const float*)d_grouped.p, d_offsets,
-                         nlist, d, (float*)ix->d_centroids.p);
-      // empty-cluster split on host (nlist x d download/upload; faiss
-      // split_clusters semantics — oracle.c states the rule)
-      std::vector<int64_t> h_off(nlist + 1);
-      (void)hipMemcpyAsync(h_off.data(), d_offsets, (nlist + 1) * 8,
-                           hipMemcpyDeviceToHost, ix->stream);
-      if (hipStreamSynchronize(ix->stream) != hipSuccess) {
-        st = DG_EINTERNAL;
-        break;
-      }
-      std::vector<int64_t> hist(nlist);
-      bool any_empty = false;
-      for (int32_t l = 0; l < nlist; l++) {
-        hist[l] = h_off[l + 1] - h_off[l];
-        if (hist[l] == 0) any_empty = true;
-      }
-      if (any_empty) {
-        (void)hipMemcpy(h_cents.data(), ix->d_centroids.p,
-                        (size_t)nlist * d * 4, hipMemcpyDeviceToHost);
-        const float EPS = 1.0f / 1024.0f;
-        for (int32_t ci = 0; ci < nlist; ci++) {
-          if (hist[ci] != 0) continue;
-          int32_t cj = 0;
-          for (;; cj = (cj + 1) % nlist) {
-            float p = (hist[cj] - 1.0f) / (float)(nt - nlist);
-            if (split_rng.rand_float() < p) break;
-          }
-          memcpy(&h_cents[(size_t)ci * d], &h_cents[(size_t)cj * d],
-                 (size_t)d * 4);
-          for (int32_t j = 0; j < d; j++) {
-            float s = (j % 2 == 0) ? 1 + EPS : 1 - EPS;
-            h_cents[(size_t)ci * d + j] *= s;
-            h_cents[(size_t)cj * d + j] *= 2 - s;
-          }
-          hist[ci] = hist[cj] / 2;
-          hist[cj] -= hist[ci];
-        }
-        (void)hipMemcpy(ix->d_centroids.p, h_cents.data(),
-                        (size_t)nlist * d * 4, hipMemcpyHostToDevice);
-      }
-    }
+      dgk::normalize_rows(ix->stream, (float*)d_td.p, n, d);
+    if ((st = dbuf_reserve(ix->d_centroids, (size_t)nlist * d * 4, ix->stream,
+                           false)) != DG_OK ||
+        (st = dbuf_reserve(ix->d_cnorms, (size_t)nlist * 4, ix->stream,
+                           false)) != DG_OK)
+      break;
+    st = kmeans_device(ix, (const float*)d_td.p, n, d, nlist,
+                       ix->desc.metric, seed, (float*)ix->d_centroids.p);
     if (st != DG_OK) break;
     dgk::row_norms(ix->stream, (const float*)ix->d_centroids.p, nlist, d,
                    (float*)ix->d_cnorms.p);
+
+    if (ix->desc.kind == DG_INDEX_IVF_PQ) {
+      // PQ encoder training (faiss IndexIVFPQ::train_encoder restated):
+      // per-subspace 256-centroid L2 k-means over coarse residuals of the
+      // train set (kmeans_device subsamples to 256*256 internally).
+      const int32_t M = ix->desc.pq_m;
+      const int32_t dsub = d / M;
+      dg_dbuf d_asg{}, d_res{}, d_sub{};
+      if ((st = dbuf_reserve(d_asg, (size_t)n * 4, ix->stream, false)) !=
+              DG_OK ||
+          (st = dbuf_reserve(d_res, (size_t)n * d * 4, ix->stream, false)) !=
+              DG_OK ||
+          (st = dbuf_reserve(d_sub, (size_t)n * dsub * 4, ix->stream,
+                             false)) != DG_OK ||
+          (st = dbuf_reserve(ix->d_codebooks,
+                             (size_t)M * 256 * dsub * 4, ix->stream,
+                             false)) != DG_OK) {
+        dbuf_free(d_asg);
+        dbuf_free(d_res);
+        dbuf_free(d_sub);
+        break;
+      }
+      st = assign_rows(ix, (const float*)d_td.p, n, (int32_t*)d_asg.p);
+      if (st == DG_OK) {
+        dgk::residual(ix->stream, (const float*)d_td.p,
+                      (const int32_t*)d_asg.p,
+                      (const float*)ix->d_centroids.p, n, d,
+                      (float*)d_res.p);
+        for (int32_t m = 0; m < M && st == DG_OK; m++) {
+          // strided column extract: subspace m of every residual row
+          if (hipMemcpy2DAsync(d_sub.p, (size_t)dsub * 4,
+                               (const char*)d_res.p + (size_t)m * dsub * 4,
+                               (size_t)d * 4, (size_t)dsub * 4, (size_t)n,
+                               hipMemcpyDeviceToDevice,
+                               ix->stream) != hipSuccess) {
+            st = DG_EINTERNAL;
+            break;
+          }
+          st = kmeans_device(ix, (const float*)d_sub.p, n, dsub, 256,
+                             DG_METRIC_L2, seed,
+                             (float*)ix->d_codebooks.p +
+                                 (size_t)m * 256 * dsub);
+        }
+      }
+      dbuf_free(d_asg);
+      dbuf_free(d_res);
+      dbuf_free(d_sub);
+      if (st != DG_OK) break;
+      st = finish_pq_tables(ix);
+      if (st != DG_OK) break;
+      ix->pq_trained = true;
+    }
     if (hipStreamSynchronize(ix->stream) != hipSuccess) st = DG_EINTERNAL;
   } while (0);
   dbuf_free(d_td);
-  dbuf_free(d_grouped);
-  dbuf_free(d_asg);
-  dbuf_free(d_perm);
-  dbuf_free(d_off);
   if (st == DG_OK) {
     ix->trained = true;
     ix->csr_valid = false;
   }
+  return st;
+}
+
+// PQ encode: residual -> per-subspace argmin against codebooks.
+// d_x must already be cosine-normalized when applicable.
+static dg_status pq_encode(dg_index* ix, const float* d_x, int64_t n,
+                           const int32_t* d_assign, uint8_t* d_codes_out) {
+  const int32_t M = ix->desc.pq_m;
+  const int32_t d = ix->desc.d;
+  const int32_t dsub = d / M;
+  const int64_t CH = std::min<int64_t>(n, 1 << 20);
+  dg_dbuf d_res{}, d_dots{}, d_amin{};
+  dg_status st;
+  if ((st = dbuf_reserve(d_res, (size_t)CH * d * 4, ix->stream, false)) !=
+          DG_OK ||
+      (st = dbuf_reserve(d_dots, (size_t)CH * 256 * 4, ix->stream, false)) !=
+          DG_OK ||
+      (st = dbuf_reserve(d_amin, (size_t)CH * 4, ix->stream, false)) !=
+          DG_OK) {
+    dbuf_free(d_res);
+    dbuf_free(d_dots);
+    dbuf_free(d_amin);
+    return st;
+  }
+  const float one = 1.0f, zero = 0.0f;
+  for (int64_t s0 = 0; s0 < n && st == DG_OK; s0 += CH) {
+    int64_t c = std::min(CH, n - s0);
+    dgk::residual(ix->stream, d_x + (size_t)s0 * d, d_assign + s0,
+                  (const float*)ix->d_centroids.p, c, d, (float*)d_res.p);
+    for (int32_t m = 0; m < M; m++) {
+      if (rocblas_sgemm(
+              ix->blas, rocblas_operation_transpose, rocblas_operation_none,
+              256, (rocblas_int)c, dsub, &one,
+              (const float*)ix->d_codebooks.p + (size_t)m * 256 * dsub, dsub,
+              (const float*)d_res.p + m * dsub, d, &zero, (float*)d_dots.p,
+              256) != rocblas_status_success) {
+        dg_set_error("pq encode sgemm failed");
+        st = DG_EINTERNAL;
+        break;
+      }
+      dgk::argmin_rows(ix->stream, (const float*)d_dots.p,
+                       (const float*)ix->d_cb_norms.p + (size_t)m * 256, c,
+                       256, DG_METRIC_L2, (int32_t*)d_amin.p);
+      dgk::set_code(ix->stream, (const int32_t*)d_amin.p, c, m, M,
+                    d_codes_out + (size_t)s0 * M);
+    }
+  }
+  dbuf_free(d_res);
+  dbuf_free(d_dots);
+  dbuf_free(d_amin);
   return st;
 }
 
@@ -481,14 +676,33 @@ static dg_status add_impl(dg_index* ix, int64_t n, const int64_t* ids,
   DeviceGuard g(ix->device);
   const int32_t d = ix->desc.d;
   const int64_t n0 = ix->ntotal;
-  dg_status st =
-      dbuf_reserve(ix->d_vectors, (size_t)(n0 + n) * d * 4, ix->stream, true);
+  const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
+  if (is_pq && !ix->pq_trained) {
+    dg_set_error("PQ encoder not trained (EVECTOR_NOT_TRAIN)");
+    return DG_ENOT_TRAINED;
+  }
+  dg_status st = DG_OK;
+  if (!is_pq)  // PQ never keeps raw vectors (cfg D: 100M x 768 > HBM)
+    st = dbuf_reserve(ix->d_vectors, (size_t)(n0 + n) * d * 4, ix->stream,
+                      true);
   if (st == DG_OK)
     st = dbuf_reserve(ix->d_ids, (size_t)(n0 + n) * 8, ix->stream, true);
   if (st == DG_OK)
     st = dbuf_reserve(ix->d_assign, (size_t)(n0 + n) * 4, ix->stream, true);
+  if (st == DG_OK && is_pq)
+    st = dbuf_reserve(ix->d_codes, (size_t)(n0 + n) * ix->desc.pq_m,
+                      ix->stream, true);
   if (st != DG_OK) return st;
-  float* dst = (float*)ix->d_vectors.p + (size_t)n0 * d;
+  dg_dbuf staging{};
+  float* dst;
+  if (is_pq) {  // stage the incoming chunk (never mutate the caller's)
+    if ((st = dbuf_reserve(staging, (size_t)n * d * 4, ix->stream, false)) !=
+        DG_OK)
+      return st;
+    dst = (float*)staging.p;
+  } else {
+    dst = (float*)ix->d_vectors.p + (size_t)n0 * d;
+  }
   DG_HIP_CHECK(hipMemcpyAsync(
       dst, x, (size_t)n * d * 4,
       x_on_device ? hipMemcpyDeviceToDevice : hipMemcpyHostToDevice,
@@ -497,13 +711,20 @@ static dg_status add_impl(dg_index* ix, int64_t n, const int64_t* ids,
                               hipMemcpyHostToDevice, ix->stream));
   if (ix->desc.metric == DG_METRIC_COSINE)
     dgk::normalize_rows(ix->stream, dst, n, d);
-  if (ix->desc.kind == DG_INDEX_IVF_FLAT) {
-    st = assign_rows(ix, dst, n, (int32_t*)ix->d_assign.p + n0);
-    if (st != DG_OK) return st;
-  } else {
+  if (ix->desc.kind == DG_INDEX_FLAT) {
     dgk::iota_i32(ix->stream, (int32_t*)ix->d_assign.p + n0, n, 0);
+  } else {
+    st = assign_rows(ix, dst, n, (int32_t*)ix->d_assign.p + n0);
+    if (st == DG_OK && is_pq)
+      st = pq_encode(ix, dst, n, (const int32_t*)ix->d_assign.p + n0,
+                     (uint8_t*)ix->d_codes.p + (size_t)n0 * ix->desc.pq_m);
+    if (st != DG_OK) {
+      dbuf_free(staging);
+      return st;
+    }
   }
   DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  dbuf_free(staging);
   for (int64_t i = 0; i < n; i++) ix->id_count.emplace(ids[i], 1);
   ix->ntotal += n;
   ix->csr_valid = false;
@@ -607,7 +828,8 @@ extern "C" dg_status dg_set_list_mask(dg_index* ix, const uint8_t* mask) {
 // ---------------- finalize: arrival arrays -> CSR ----------------
 static dg_status finalize_csr(dg_index* ix) {
   const bool is_ivf = ix->desc.kind == DG_INDEX_IVF_FLAT;
-  const int32_t nlist = is_ivf ? ix->desc.nlist : 1;
+  const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
+  const int32_t nlist = ix->desc.kind == DG_INDEX_FLAT ? 1 : ix->desc.nlist;
   const int32_t d = ix->desc.d;
   const int64_t n = ix->ntotal;
   const int32_t CR = dg_index::kChunkRows;
@@ -617,18 +839,24 @@ static dg_status finalize_csr(dg_index* ix) {
   dg_dbuf& rowmajor = is_ivf ? rm_tmp : ix->d_csr_vectors;
   if ((st = dbuf_reserve(ix->d_csr_offsets, ((size_t)nlist + 1) * 8,
                          ix->stream, false)) != DG_OK ||
-      (st = dbuf_reserve(rowmajor, (size_t)n * d * 4, ix->stream, false)) !=
-          DG_OK ||
       (st = dbuf_reserve(ix->d_csr_ids, (size_t)n * 8, ix->stream, false)) !=
           DG_OK ||
-      (st = dbuf_reserve(ix->d_csr_vnorms, (size_t)n * 4, ix->stream,
-                         false)) != DG_OK ||
       (st = dbuf_reserve(ix->ws_small,
                          (size_t)nlist * 4 + (size_t)n * 4 + 64, ix->stream,
-                         false)) != DG_OK) {
+                         false)) != DG_OK)
+    return st;
+  if (!is_pq &&
+      ((st = dbuf_reserve(rowmajor, (size_t)n * d * 4, ix->stream, false)) !=
+           DG_OK ||
+       (st = dbuf_reserve(ix->d_csr_vnorms, (size_t)n * 4, ix->stream,
+                          false)) != DG_OK)) {
     dbuf_free(rm_tmp);
     return st;
   }
+  if (is_pq && (st = dbuf_reserve(ix->d_csr_codes,
+                                  (size_t)n * ix->desc.pq_m, ix->stream,
+                                  false)) != DG_OK)
+    return st;
   int32_t* d_counts = (int32_t*)ix->ws_small.p;
   uint32_t* d_perm = (uint32_t*)((char*)ix->ws_small.p + (size_t)nlist * 4);
   int64_t* d_offsets = (int64_t*)ix->d_csr_offsets.p;
@@ -640,12 +868,17 @@ static dg_status finalize_csr(dg_index* ix) {
     dgk::init_cursors(ix->stream, d_offsets, nlist, d_counts);
     dgk::scatter_perm(ix->stream, (const int32_t*)ix->d_assign.p, n, nullptr,
                       d_counts, d_perm);
-    dgk::gather_rows(ix->stream, (const float*)ix->d_vectors.p, d_perm, n, d,
-                     (float*)rowmajor.p);
+    if (is_pq) {
+      dgk::gather_codes(ix->stream, (const uint8_t*)ix->d_codes.p, d_perm, n,
+                        ix->desc.pq_m, (uint8_t*)ix->d_csr_codes.p);
+    } else {
+      dgk::gather_rows(ix->stream, (const float*)ix->d_vectors.p, d_perm, n,
+                       d, (float*)rowmajor.p);
+      dgk::row_norms(ix->stream, (const float*)rowmajor.p, n, d,
+                     (float*)ix->d_csr_vnorms.p);
+    }
     dgk::gather_ids(ix->stream, (const int64_t*)ix->d_ids.p, d_perm, n,
                     (int64_t*)ix->d_csr_ids.p);
-    dgk::row_norms(ix->stream, (const float*)rowmajor.p, n, d,
-                   (float*)ix->d_csr_vnorms.p);
   } else {
     (void)hipMemsetAsync(d_offsets, 0, ((size_t)nlist + 1) * 8, ix->stream);
   }
@@ -718,7 +951,8 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                              int64_t* d_out_ids) {
   const int32_t d = ix->desc.d;
   const int metric = ix->desc.metric;
-  const bool is_ivf = ix->desc.kind == DG_INDEX_IVF_FLAT;
+  const bool is_ivf = ix->desc.kind != DG_INDEX_FLAT;
+  const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
   const int32_t nlist = is_ivf ? ix->desc.nlist : 1;
   dg_status st = DG_OK;
 
@@ -850,6 +1084,16 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     uint64_t* final_tk = (uint64_t*)(probes + (size_t)nq * np);  // nq x k
     const uint8_t* maskp =
         ix->has_mask ? (const uint8_t*)ix->d_list_mask.p : nullptr;
+    if (is_pq && np == nlist) {
+      // PQ needs the coarse dots matrix in every case (ADC bias term)
+      st = sgemm_dots(ix, dq, nq, (const float*)ix->d_centroids.p, nlist, d,
+                      (float*)ix->ws_dots.p);
+      if (st != DG_OK) {
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        return st;
+      }
+    }
     if (np == nlist) {
       // full sweep: every list probed; no coarse selection needed (also the
       // exact-ground-truth path recall measurement uses)
@@ -939,12 +1183,14 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
     int64_t total_cand = totals[0];
     int32_t total_units = (int32_t)totals[1];
-    // algorithmic bytes: one read of each probed list (vectors + norms)
+    // algorithmic bytes: one read of each probed list (IVF-Flat: vectors +
+    // norms; PQ: codes — the S/T table gathers are on-chip cache traffic,
+    // DESIGN.md §ivf-pq)
     int64_t alg = 0;
+    const int64_t row_bytes = is_pq ? ix->desc.pq_m : (int64_t)(d * 4 + 4);
     for (int32_t l = 0; l < nlist; l++)
       if (h_inv_counts[l] > 0)
-        alg += (ix->h_csr_offsets[l + 1] - ix->h_csr_offsets[l]) *
-               (int64_t)(d * 4 + 4);
+        alg += (ix->h_csr_offsets[l + 1] - ix->h_csr_offsets[l]) * row_bytes;
     ix->times.last_scan_bytes_alg = alg;
 
     size_t need = ((size_t)nlist + 1) * 8 + (size_t)total_units * 8;
@@ -962,18 +1208,52 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     dgk::fill_units(ix->stream, cursors, unit_counts, nlist, nullptr, 0,
                     units, total_units);
 
-    // THE scan (columnar v2)
-    char* mp = (char*)ix->d_chunk_meta.p;
-    const int32_t* d_chunk_off = (const int32_t*)mp;
-    const int64_t* d_chunk_base =
-        (const int64_t*)(mp + ((size_t)nlist + 1) * 4);
     (void)hipEventRecord(ix->ev[2], ix->stream);
-    dgk::ivf_scan_col(ix->stream, units, total_units,
-                      (const int64_t*)ix->d_csr_offsets.p, d_chunk_off,
-                      d_chunk_base, (const float*)ix->d_csr_t.p,
-                      (const float*)ix->d_csr_vnorms.p, dq, d, inv_offsets32,
+    if (is_pq) {
+      // ADC scan: build T (strided-batched GEMM, q_sub x codebook^T per
+      // subspace) then gather-scan codes
+      const int32_t M = ix->desc.pq_m;
+      const int32_t dsub = d / M;
+      if ((st = dbuf_reserve(ix->ws_T, (size_t)nq * M * 256 * 4, ix->stream,
+                             false)) != DG_OK) {
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        return st;
+      }
+      const float one = 1.0f, zero = 0.0f;
+      if (rocblas_sgemm_strided_batched(
+              ix->blas, rocblas_operation_transpose, rocblas_operation_none,
+              256, (rocblas_int)nq, dsub, &one,
+              (const float*)ix->d_codebooks.p, dsub, (int64_t)256 * dsub, dq,
+              d, (int64_t)dsub, &zero, (float*)ix->ws_T.p,
+              (rocblas_int)(M * 256), (int64_t)256,
+              M) != rocblas_status_success) {
+        dg_set_error("T build sgemm failed");
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        return DG_EINTERNAL;
+      }
+      dgk::ivfpq_scan(ix->stream, units, total_units,
+                      (const int64_t*)ix->d_csr_offsets.p,
+                      (const uint8_t*)ix->d_csr_codes.p,
+                      (const float*)ix->d_S.p, (const float*)ix->ws_T.p,
+                      (const float*)ix->ws_dots.p, nlist, M, inv_offsets32,
                       inv_q, inv_rank, qp_off, q_cand_base, np, metric,
                       d_bitmap, chunk_rows, (uint64_t*)ix->ws_cand.p);
+    } else {
+      // THE scan (columnar v2)
+      char* mp = (char*)ix->d_chunk_meta.p;
+      const int32_t* d_chunk_off = (const int32_t*)mp;
+      const int64_t* d_chunk_base =
+          (const int64_t*)(mp + ((size_t)nlist + 1) * 4);
+      dgk::ivf_scan_col(ix->stream, units, total_units,
+                        (const int64_t*)ix->d_csr_offsets.p, d_chunk_off,
+                        d_chunk_base, (const float*)ix->d_csr_t.p,
+                        (const float*)ix->d_csr_vnorms.p, dq, d,
+                        inv_offsets32, inv_q, inv_rank, qp_off, q_cand_base,
+                        np, metric, d_bitmap, chunk_rows,
+                        (uint64_t*)ix->ws_cand.p);
+    }
     (void)hipEventRecord(ix->ev[3], ix->stream);
 
     // select + emit
@@ -1008,7 +1288,8 @@ extern "C" dg_status dg_search_device(dg_index* ix, int64_t nq,
   }
   DeviceGuard g(ix->device);
   // untrained IVF: blank results, OK (ivf_flat.cc:223-227)
-  if (!ix->trained || ix->ntotal == 0) {
+  if (!ix->trained || ix->ntotal == 0 ||
+      (ix->desc.kind == DG_INDEX_IVF_PQ && !ix->pq_trained)) {
     std::shared_lock lk(ix->rw);
     (void)hipMemsetAsync(d_out_dist, 0, (size_t)nq * k * 4, ix->stream);
     (void)hipMemsetAsync(d_out_ids, 0xff, (size_t)nq * k * 8, ix->stream);
@@ -1099,28 +1380,47 @@ extern "C" dg_status dg_save(dg_index* ix, const char* path) {
   fwrite(&ix->ntotal, 8, 1, f);
   fwrite(&ix->n_deleted, 8, 1, f);
   dg_status st = DG_OK;
-  if (ix->desc.kind == DG_INDEX_IVF_FLAT && ix->trained) {
+  const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
+  if (ix->desc.kind != DG_INDEX_FLAT && ix->trained) {
     std::vector<float> cents((size_t)ix->desc.nlist * d);
     if (hipMemcpy(cents.data(), ix->d_centroids.p, cents.size() * 4,
                   hipMemcpyDeviceToHost) != hipSuccess)
       st = DG_EINTERNAL;
     fwrite(cents.data(), 4, cents.size(), f);
   }
+  if (st == DG_OK && is_pq && ix->pq_trained) {
+    const int32_t dsub = d / ix->desc.pq_m;
+    std::vector<float> cb((size_t)ix->desc.pq_m * 256 * dsub);
+    if (hipMemcpy(cb.data(), ix->d_codebooks.p, cb.size() * 4,
+                  hipMemcpyDeviceToHost) != hipSuccess)
+      st = DG_EINTERNAL;
+    fwrite(cb.data(), 4, cb.size(), f);
+  }
   if (st == DG_OK && ix->ntotal > 0) {
     const size_t CH = 1 << 20;  // rows per host staging chunk
-    std::vector<float> vbuf(CH * d);
+    const int32_t M = ix->desc.pq_m;
+    std::vector<float> vbuf(is_pq ? 0 : CH * d);
+    std::vector<uint8_t> cbuf(is_pq ? CH * M : 0);
     std::vector<int64_t> ibuf(CH);
     std::vector<int32_t> abuf(CH);
     for (int64_t s0 = 0; s0 < ix->ntotal && st == DG_OK; s0 += CH) {
       size_t c = std::min<int64_t>(CH, ix->ntotal - s0);
-      if (hipMemcpy(vbuf.data(), (float*)ix->d_vectors.p + (size_t)s0 * d,
-                    c * d * 4, hipMemcpyDeviceToHost) != hipSuccess ||
-          hipMemcpy(ibuf.data(), (int64_t*)ix->d_ids.p + s0, c * 8,
+      if (is_pq) {
+        if (hipMemcpy(cbuf.data(), (uint8_t*)ix->d_codes.p + (size_t)s0 * M,
+                      c * M, hipMemcpyDeviceToHost) != hipSuccess)
+          st = DG_EINTERNAL;
+        fwrite(cbuf.data(), 1, c * M, f);
+      } else {
+        if (hipMemcpy(vbuf.data(), (float*)ix->d_vectors.p + (size_t)s0 * d,
+                      c * d * 4, hipMemcpyDeviceToHost) != hipSuccess)
+          st = DG_EINTERNAL;
+        fwrite(vbuf.data(), 4, c * d, f);
+      }
+      if (hipMemcpy(ibuf.data(), (int64_t*)ix->d_ids.p + s0, c * 8,
                     hipMemcpyDeviceToHost) != hipSuccess ||
           hipMemcpy(abuf.data(), (int32_t*)ix->d_assign.p + s0, c * 4,
                     hipMemcpyDeviceToHost) != hipSuccess)
         st = DG_EINTERNAL;
-      fwrite(vbuf.data(), 4, c * d, f);
       fwrite(ibuf.data(), 8, c, f);
       fwrite(abuf.data(), 4, c, f);
     }
@@ -1157,8 +1457,9 @@ extern "C" dg_status dg_load(dg_index** out, const char* path,
   }
   DeviceGuard g(ix->device);
   const int32_t d = desc.d;
+  const bool is_pq = desc.kind == DG_INDEX_IVF_PQ;
   do {
-    if (desc.kind == DG_INDEX_IVF_FLAT && trained) {
+    if (desc.kind != DG_INDEX_FLAT && trained) {
       std::vector<float> cents((size_t)desc.nlist * d);
       if (fread(cents.data(), 4, cents.size(), f) != cents.size()) {
         st = DG_EIO;
@@ -1167,29 +1468,61 @@ extern "C" dg_status dg_load(dg_index** out, const char* path,
       st = dg_set_centroids(ix, desc.nlist, cents.data());
       if (st != DG_OK) break;
     }
+    if (is_pq && trained) {
+      const int32_t dsub = d / desc.pq_m;
+      std::vector<float> cb((size_t)desc.pq_m * 256 * dsub);
+      if (fread(cb.data(), 4, cb.size(), f) != cb.size()) {
+        st = DG_EIO;
+        break;
+      }
+      st = dg_set_codebooks(ix, desc.pq_m, 8, cb.data());
+      if (st != DG_OK) break;
+    }
     if (ntotal > 0) {
-      if ((st = dbuf_reserve(ix->d_vectors, (size_t)ntotal * d * 4,
-                             ix->stream, false)) != DG_OK ||
-          (st = dbuf_reserve(ix->d_ids, (size_t)ntotal * 8, ix->stream,
+      const int32_t M = desc.pq_m;
+      if ((st = dbuf_reserve(ix->d_ids, (size_t)ntotal * 8, ix->stream,
                              false)) != DG_OK ||
           (st = dbuf_reserve(ix->d_assign, (size_t)ntotal * 4, ix->stream,
                              false)) != DG_OK)
         break;
+      if (!is_pq && (st = dbuf_reserve(ix->d_vectors, (size_t)ntotal * d * 4,
+                                       ix->stream, false)) != DG_OK)
+        break;
+      if (is_pq && (st = dbuf_reserve(ix->d_codes, (size_t)ntotal * M,
+                                      ix->stream, false)) != DG_OK)
+        break;
       const size_t CH = 1 << 20;
-      std::vector<float> vbuf(CH * d);
+      std::vector<float> vbuf(is_pq ? 0 : CH * d);
+      std::vector<uint8_t> cbuf(is_pq ? CH * M : 0);
       std::vector<int64_t> ibuf(CH);
       std::vector<int32_t> abuf(CH);
       for (int64_t s0 = 0; s0 < ntotal && st == DG_OK; s0 += CH) {
         size_t c = std::min<int64_t>(CH, ntotal - s0);
-        if (fread(vbuf.data(), 4, c * d, f) != c * d ||
-            fread(ibuf.data(), 8, c, f) != c ||
+        if (is_pq) {
+          if (fread(cbuf.data(), 1, c * M, f) != c * M) {
+            st = DG_EIO;
+            break;
+          }
+          if (hipMemcpy((uint8_t*)ix->d_codes.p + (size_t)s0 * M,
+                        cbuf.data(), c * M,
+                        hipMemcpyHostToDevice) != hipSuccess)
+            st = DG_EINTERNAL;
+        } else {
+          if (fread(vbuf.data(), 4, c * d, f) != c * d) {
+            st = DG_EIO;
+            break;
+          }
+          if (hipMemcpy((float*)ix->d_vectors.p + (size_t)s0 * d,
+                        vbuf.data(), c * d * 4,
+                        hipMemcpyHostToDevice) != hipSuccess)
+            st = DG_EINTERNAL;
+        }
+        if (fread(ibuf.data(), 8, c, f) != c ||
             fread(abuf.data(), 4, c, f) != c) {
           st = DG_EIO;
           break;
         }
-        if (hipMemcpy((float*)ix->d_vectors.p + (size_t)s0 * d, vbuf.data(),
-                      c * d * 4, hipMemcpyHostToDevice) != hipSuccess ||
-            hipMemcpy((int64_t*)ix->d_ids.p + s0, ibuf.data(), c * 8,
+        if (hipMemcpy((int64_t*)ix->d_ids.p + s0, ibuf.data(), c * 8,
                       hipMemcpyHostToDevice) != hipSuccess ||
             hipMemcpy((int32_t*)ix->d_assign.p + s0, abuf.data(), c * 4,
                       hipMemcpyHostToDevice) != hipSuccess)

@@ -100,6 +100,16 @@ struct dg_index {
   // Flat uses d_csr_* with a single implicit list (nlist=1) so scan/select
   // machinery is shared.
 
+  // ---- IVF-PQ state ----
+  bool pq_trained = false;
+  dg_dbuf d_codebooks;      // [M][256][dsub] f32
+  dg_dbuf d_codes;          // arrival [ntotal x M] u8 (raw vectors are NOT
+                            // kept for PQ; d_vectors stays empty)
+  dg_dbuf d_csr_codes;      // grouped [ntotal x M] u8
+  dg_dbuf d_S;              // [nlist][M][256] f32 ||c_m + cb||^2
+  dg_dbuf d_cb_norms;       // [M][256] f32 codebook entry norms (encode)
+  dg_dbuf ws_T;             // per-batch [nq][M][256] f32 q_sub . cb
+
   // optional list ownership mask (multi-GPU list sharding)
   dg_dbuf d_list_mask;      // [nlist] u8, empty = all owned
   bool has_mask = false;
@@ -189,6 +199,23 @@ void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
                       const int64_t* csr_offsets, const int32_t* chunk_off,
                       const int64_t* chunk_base, const float* rowmajor,
                       int32_t d, int32_t chunk_rows, float* tvec);
+// IVF-PQ
+void residual(hipStream_t s, const float* x, const int32_t* assign,
+              const float* centroids, int64_t n, int32_t d, float* out);
+void set_code(hipStream_t s, const int32_t* amin, int64_t n, int32_t m,
+              int32_t M, uint8_t* codes);
+void gather_codes(hipStream_t s, const uint8_t* src, const uint32_t* perm,
+                  int64_t n, int32_t M, uint8_t* dst);
+void build_S(hipStream_t s, const float* centroids, const float* codebooks,
+             int32_t nlist, int32_t M, int32_t dsub, int32_t d, float* S);
+void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
+                const int64_t* csr_offsets, const uint8_t* csr_codes,
+                const float* S, const float* T, const float* coarse_dots,
+                int32_t nlist, int32_t M, const int32_t* inv_offsets,
+                const int32_t* inv_q, const int32_t* inv_rank,
+                const int64_t* qp_off, const int64_t* q_cand_base,
+                int32_t nprobe, int metric, const uint32_t* bitmap,
+                int32_t chunk_rows, uint64_t* cand);
 // emit: resolve ids, apply metric convention
 void emit_results(hipStream_t s, const uint64_t* topk,
                   const int64_t* ids_lookup, const float* qnorms, int64_t nq,

@@ -646,6 +646,138 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
   }
 }
 
+// ---------- IVF-PQ kernels ----------
+// ADC decomposition: for L2,
+//   ||q - (c_l + r)||^2 = ||q||^2 - 2*dots[q][l]
+//                         + sum_m ( S[l][m][code] - 2*T[q][m][code] )
+// with S[l][m][c] = ||centroid_l_sub_m + codebook_m[c]||^2 (index-static)
+// and  T[q][m][c] = q_sub_m . codebook_m[c] (per batch, strided GEMM).
+// For IP: score = dots[q][l] + sum_m T[q][m][code].
+// Restates faiss IndexIVFPQ residual ADC (oracle.c dgo_ivfpq_search is the
+// direct restatement; this is the same arithmetic regrouped).
+
+__global__ void k_residual(const float* __restrict__ x,
+                           const int32_t* __restrict__ assign,
+                           const float* __restrict__ centroids, int64_t n,
+                           int32_t d, float* __restrict__ out) {
+  int64_t row = (int64_t)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  if (row >= n) return;
+  const float* v = x + (size_t)row * d;
+  const float* c = centroids + (size_t)assign[row] * d;
+  float* o = out + (size_t)row * d;
+  for (int i = lane; i < d; i += WAVE) o[i] = v[i] - c[i];
+}
+
+__global__ void k_set_code(const int32_t* __restrict__ amin, int64_t n,
+                           int32_t m, int32_t M, uint8_t* __restrict__ codes) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) codes[(size_t)i * M + m] = (uint8_t)amin[i];
+}
+
+__global__ void k_gather_codes(const uint8_t* __restrict__ src,
+                               const uint32_t* __restrict__ perm, int64_t n,
+                               int32_t M, uint8_t* __restrict__ dst) {
+  // dst[perm[i]] = src[i], M bytes per row, dword copies
+  int64_t i = (int64_t)blockIdx.x * (blockDim.x / 32) + threadIdx.x / 32;
+  int sub = threadIdx.x % 32;
+  if (i >= n) return;
+  const uint32_t* s = (const uint32_t*)(src + (size_t)i * M);
+  uint32_t* t = (uint32_t*)(dst + (size_t)perm[i] * M);
+  for (int j = sub; j < M / 4; j += 32) t[j] = s[j];
+}
+
+__global__ void k_build_S(const float* __restrict__ centroids,
+                          const float* __restrict__ codebooks, int32_t nlist,
+                          int32_t M, int32_t dsub, int32_t d,
+                          float* __restrict__ S) {
+  // one thread per (l, m, code): ||c_sub + cb||^2 over dsub elems
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)nlist * M * 256;
+  if (i >= total) return;
+  int32_t code = (int32_t)(i & 255);
+  int32_t m = (int32_t)((i >> 8) % M);
+  int32_t l = (int32_t)(i / (256 * M));
+  const float* c = centroids + (size_t)l * d + m * dsub;
+  const float* cb = codebooks + ((size_t)m * 256 + code) * dsub;
+  float acc = 0.f;
+  for (int32_t j = 0; j < dsub; j++) {
+    float t = c[j] + cb[j];
+    acc += t * t;
+  }
+  S[i] = acc;
+}
+
+// THE PQ scan: unit = (list, chunk of codes).  Codes for a 256-vector tile
+// are LDS-staged once and reused by every probing query; each LANE owns one
+// vector and walks m = 0..M-1 sequentially, so all lanes of a wave gather
+// within the same 1 KB rows of T[q] and S[l] (L1-resident after first
+// touch).  No cross-lane reduction.
+__global__ void __launch_bounds__(256, 4) k_ivfpq_scan(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const uint8_t* __restrict__ csr_codes, const float* __restrict__ S,
+    const float* __restrict__ T, const float* __restrict__ coarse_dots,
+    int32_t nlist, int32_t M, const int32_t* __restrict__ inv_offsets,
+    const int32_t* __restrict__ inv_q, const int32_t* __restrict__ inv_rank,
+    const int64_t* __restrict__ qp_off, const int64_t* __restrict__ q_cand_base,
+    int32_t nprobe, int metric, const uint32_t* __restrict__ bitmap,
+    int32_t chunk_rows, uint64_t* __restrict__ cand) {
+  extern __shared__ __attribute__((aligned(16))) uint8_t lds_codes[];
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t list_end = csr_offsets[list + 1];
+  const int64_t row_start = list_start + (int64_t)chunk * chunk_rows;
+  const int64_t row_end = min(list_end, row_start + chunk_rows);
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const float* Sl = S + (size_t)list * M * 256;
+
+  const int32_t TILE = 256;  // vectors staged per pass (TILE*M bytes LDS)
+  for (int64_t t0 = row_start; t0 < row_end; t0 += TILE) {
+    const int32_t tn = (int32_t)min((int64_t)TILE, row_end - t0);
+    __syncthreads();
+    {  // stage codes coalesced (rows are contiguous in csr_codes)
+      const uint32_t* src = (const uint32_t*)(csr_codes + (size_t)t0 * M);
+      uint32_t* dst = (uint32_t*)lds_codes;
+      int32_t words = tn * M / 4;
+      for (int32_t w = threadIdx.x; w < words; w += blockDim.x)
+        dst[w] = src[w];
+    }
+    __syncthreads();
+
+    // waves split the probing queries; lanes split the tile's vectors
+    for (int32_t qi = wave_id; qi < nql; qi += blockDim.x / WAVE) {
+      const int32_t q = inv_q[iq0 + qi];
+      const int32_t rank = inv_rank[iq0 + qi];
+      const float* Tq = T + (size_t)q * M * 256;
+      const float dot = coarse_dots[(size_t)q * nlist + list];
+      const int64_t cb0 = q_cand_base[q] +
+                          qp_off[(int64_t)q * nprobe + rank] - list_start;
+      for (int32_t v = lane; v < tn; v += WAVE) {
+        const uint8_t* code = lds_codes + (size_t)v * M;
+        float acc = 0.f;
+        if (metric == 0) {
+          for (int32_t m = 0; m < M; m++) {
+            const int32_t c = code[m];
+            acc += Sl[m * 256 + c] - 2.0f * Tq[m * 256 + c];
+          }
+          acc -= 2.0f * dot;  // + qnorm at emit
+        } else {
+          for (int32_t m = 0; m < M; m++) acc += Tq[m * 256 + code[m]];
+          acc = -(acc + dot);  // IP key = -score
+        }
+        const int64_t r = t0 + v;
+        bool pass = true;
+        if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+        cand[cb0 + r] = pass ? pack_cand(acc, (uint32_t)r) : kCandEmpty;
+      }
+    }
+  }
+}
+
 // tiled row-major -> column-major chunk transpose (finalize step).
 // One block per (list, chunk) unit over ALL chunks; LDS 64x65 tile.
 __global__ void k_transpose_chunks(
@@ -996,6 +1128,47 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     default: DG_SCAN_LAUNCH(16, 4); break;
   }
 #undef DG_SCAN_LAUNCH
+}
+
+void residual(hipStream_t s, const float* x, const int32_t* assign,
+              const float* centroids, int64_t n, int32_t d, float* out) {
+  if (n) hipLaunchKernelGGL(k_residual, dim3(ceil_div(n, 4)), dim3(4 * WAVE),
+                            0, s, x, assign, centroids, n, d, out);
+}
+
+void set_code(hipStream_t s, const int32_t* amin, int64_t n, int32_t m,
+              int32_t M, uint8_t* codes) {
+  if (n) hipLaunchKernelGGL(k_set_code, dim3(ceil_div(n, 256)), dim3(256), 0,
+                            s, amin, n, m, M, codes);
+}
+
+void gather_codes(hipStream_t s, const uint8_t* src, const uint32_t* perm,
+                  int64_t n, int32_t M, uint8_t* dst) {
+  if (n) hipLaunchKernelGGL(k_gather_codes, dim3(ceil_div(n, 8)),
+                            dim3(8 * 32), 0, s, src, perm, n, M, dst);
+}
+
+void build_S(hipStream_t s, const float* centroids, const float* codebooks,
+             int32_t nlist, int32_t M, int32_t dsub, int32_t d, float* S) {
+  int64_t total = (int64_t)nlist * M * 256;
+  hipLaunchKernelGGL(k_build_S, dim3(ceil_div(total, 256)), dim3(256), 0, s,
+                     centroids, codebooks, nlist, M, dsub, d, S);
+}
+
+void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
+                const int64_t* csr_offsets, const uint8_t* csr_codes,
+                const float* S, const float* T, const float* coarse_dots,
+                int32_t nlist, int32_t M, const int32_t* inv_offsets,
+                const int32_t* inv_q, const int32_t* inv_rank,
+                const int64_t* qp_off, const int64_t* q_cand_base,
+                int32_t nprobe, int metric, const uint32_t* bitmap,
+                int32_t chunk_rows, uint64_t* cand) {
+  if (!n_units) return;
+  size_t lds = 256 * (size_t)M;
+  hipLaunchKernelGGL(k_ivfpq_scan, dim3((uint32_t)n_units), dim3(256), lds, s,
+                     units, csr_offsets, csr_codes, S, T, coarse_dots, nlist,
+                     M, inv_offsets, inv_q, inv_rank, qp_off, q_cand_base,
+                     nprobe, metric, bitmap, chunk_rows, cand);
 }
 
 void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,

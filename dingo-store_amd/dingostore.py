@@ -139,14 +139,16 @@ def make_filter(kind=0, negate=False, min_id=0, max_id=0, ids=None,
 
 
 class Index:
-    def __init__(self, kind, metric, d, nlist=0, device=-1, _handle=None):
+    def __init__(self, kind, metric, d, nlist=0, m=0, device=-1,
+                 reserve=0, _handle=None):
         self.kind, self.metric, self.d = kind, metric, d
         self.nlist = nlist
+        self.m = m
         if _handle is not None:
             self.h = _handle
             return
         desc = _Desc(kind=kind, metric=metric, d=d, nlist=nlist,
-                     pq_m=0, pq_nbits=8, device=device, reserve=0)
+                     pq_m=m, pq_nbits=8, device=device, reserve=reserve)
         h = C.c_void_p()
         _check(lib().dg_index_create(C.byref(h), C.byref(desc)),
                "dg_index_create")
@@ -176,6 +178,25 @@ class Index:
     def get_centroids(self):
         out = np.empty((self.nlist, self.d), np.float32)
         _check(lib().dg_get_centroids(self.h, out), "dg_get_centroids")
+        return out
+
+    def set_codebooks(self, codebooks):
+        """codebooks: [m, 256, d//m] float32."""
+        cb = np.ascontiguousarray(codebooks, np.float32)
+        self.m = cb.shape[0]
+        l = lib()
+        l.dg_set_codebooks.argtypes = [C.c_void_p, C.c_int32, C.c_int32,
+                                       _f32p]
+        _check(l.dg_set_codebooks(self.h, cb.shape[0], 8, cb.reshape(-1)),
+               "dg_set_codebooks")
+
+    def get_codebooks(self):
+        dsub = self.d // self.m
+        out = np.empty((self.m, 256, dsub), np.float32)
+        l = lib()
+        l.dg_get_codebooks.argtypes = [C.c_void_p, _f32p]
+        _check(l.dg_get_codebooks(self.h, out.reshape(-1)),
+               "dg_get_codebooks")
         return out
 
     def add(self, ids, x):

@@ -143,6 +143,12 @@ dg_status dg_train(dg_index* idx, int64_t n, const float* x);
  * (BASELINE.md protocol).  set_centroids marks the index trained. */
 dg_status dg_set_centroids(dg_index* idx, int32_t nlist, const float* centroids);
 dg_status dg_get_centroids(dg_index* idx, float* out_centroids);
+/* IVF-PQ codebook injection/extraction (m x 256 x (d/m) floats), the PQ
+ * analog of dg_set_centroids for oracle-shared parity (requires centroids
+ * set first; rebuilds the ADC tables). */
+dg_status dg_set_codebooks(dg_index* idx, int32_t m, int32_t nbits,
+                           const float* codebooks);
+dg_status dg_get_codebooks(dg_index* idx, float* out_codebooks);
 
 /* ---- mutation (exclusive) ----
  * add restates faiss add_with_ids via VectorIndexIvfFlat::Add

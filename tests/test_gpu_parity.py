@@ -374,3 +374,98 @@ def test_stats_timing_populated():
     assert st["last_nq"] == 64
     assert st["last_scan_ms"] > 0
     assert st["last_scan_bytes_algorithmic"] > 0
+
+
+# ---------------- IVF-PQ ----------------
+def build_pq_pair(metric, base, nlist, m):
+    """GPU PQ index + oracle PQ structure sharing centroids AND codebooks."""
+    n, d = base.shape
+    ob = base.copy()
+    if metric == orc.COSINE:
+        ob = orc.normalize(ob)
+    cents = orc.kmeans(metric, ob, nlist)
+    assign = orc.ivf_assign(metric, ob, cents)
+    residuals = ob - cents[assign]
+    cb = orc.pq_train(residuals, m)
+    gpu = dg.Index(dg.IVF_PQ, metric, d, nlist=nlist, m=m)
+    gpu.set_centroids(cents)
+    gpu.set_codebooks(cb)
+    gpu.add(np.arange(n, dtype=np.int64), base)
+    codes = orc.ivfpq_encode(ob, assign, cents, cb)
+    off, _, gi = orc.ivf_build(ob, None, nlist, assign)
+    gcodes = np.empty_like(codes)
+    cursor = off[:-1].copy()
+    for i in range(n):
+        gcodes[cursor[assign[i]]] = codes[i]
+        cursor[assign[i]] += 1
+    return gpu, (cents, cb, off, gcodes, gi)
+
+
+@pytest.mark.parametrize("metric", [orc.L2, orc.IP])
+def test_ivfpq_parity(metric):
+    base, q = make_data(n=20000, d=64, nq=64)
+    nlist, m, nprobe, k = 32, 8, 8, 10
+    gpu, (cents, cb, off, gcodes, gi_) = build_pq_pair(metric, base, nlist, m)
+    try:
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    od, oi = orc.ivfpq_search(metric, cents, off, gcodes, gi_, cb, q, k,
+                              nprobe)
+    # ADC sums are reordered between the two implementations; compare with a
+    # slightly wider tie tolerance than exact-vector paths
+    frac = ids_match_with_tie_slack(gd, gi, od, oi, tol=5e-3)
+    assert frac >= 0.95, f"metric {metric}: {frac}"
+
+
+def test_ivfpq_gpu_train_recall():
+    """GPU-trained PQ (coarse + codebooks) recall close to oracle-trained."""
+    base, q = make_data(n=20000, d=64, nq=128)
+    nlist, m, nprobe, k = 32, 8, 8, 10
+    gt_d, gt_i = orc.flat_search(orc.L2, base, q, k)
+    gpu = dg.Index(dg.IVF_PQ, dg.L2, 64, nlist=nlist, m=m)
+    try:
+        gpu.train(base)
+        gpu.add(np.arange(base.shape[0], dtype=np.int64), base)
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    r_gpu = np.mean([len(set(a) & set(b)) / k for a, b in zip(gt_i, gi)])
+    # oracle-trained reference recall
+    gpu2, (cents, cb, off, gcodes, gi_) = build_pq_pair(orc.L2, base, nlist, m)
+    gpu2.close()
+    od, oi = orc.ivfpq_search(orc.L2, cents, off, gcodes, gi_, cb, q, k,
+                              nprobe)
+    r_orc = np.mean([len(set(a) & set(b)) / k for a, b in zip(gt_i, oi)])
+    assert r_gpu >= r_orc - 0.05, (r_gpu, r_orc)
+
+
+def test_ivfpq_save_load(tmp_path):
+    base, q = make_data(n=8000, d=64, nq=16)
+    gpu, _ = build_pq_pair(orc.L2, base, 16, 8)
+    p = str(tmp_path / "pq.dgi")
+    try:
+        gd0, gi0 = gpu.search(q, 10, nprobe=8)
+        gpu.save(p)
+    finally:
+        gpu.close()
+    idx2 = dg.Index.load(p)
+    idx2.m = 8
+    try:
+        gd1, gi1 = idx2.search(q, 10, nprobe=8)
+    finally:
+        idx2.close()
+    assert np.array_equal(gi0, gi1)
+    assert np.allclose(gd0, gd1, rtol=1e-6, atol=1e-6)
+
+
+def test_ivfpq_untrained_and_edge():
+    idx = dg.Index(dg.IVF_PQ, dg.L2, 64, nlist=16, m=8)
+    try:
+        gd, gi = idx.search(np.zeros((2, 64), np.float32), 3)
+        assert (gi == -1).all()
+        with pytest.raises(dg.DgError):
+            idx.add(np.arange(3, dtype=np.int64), np.zeros((3, 64),
+                                                           np.float32))
+    finally:
+        idx.close()
