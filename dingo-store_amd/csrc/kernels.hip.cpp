@@ -757,16 +757,34 @@ __global__ void __launch_bounds__(256, 4) k_ivfpq_scan(
       const int64_t cb0 = q_cand_base[q] +
                           qp_off[(int64_t)q * nprobe + rank] - list_start;
       for (int32_t v = lane; v < tn; v += WAVE) {
-        const uint8_t* code = lds_codes + (size_t)v * M;
+        const uint32_t* code4 = (const uint32_t*)(lds_codes + (size_t)v * M);
         float acc = 0.f;
+        // m unrolled by 4 (one LDS dword = 4 codes; 8 independent gather
+        // chains in flight instead of one serial byte->gather->add chain)
         if (metric == 0) {
-          for (int32_t m = 0; m < M; m++) {
-            const int32_t c = code[m];
-            acc += Sl[m * 256 + c] - 2.0f * Tq[m * 256 + c];
+          for (int32_t m4 = 0; m4 < M / 4; m4++) {
+            const uint32_t cw = code4[m4];
+            const int32_t m = m4 * 4;
+            float s0 = Sl[(m + 0) * 256 + (cw & 255)];
+            float s1 = Sl[(m + 1) * 256 + ((cw >> 8) & 255)];
+            float s2 = Sl[(m + 2) * 256 + ((cw >> 16) & 255)];
+            float s3 = Sl[(m + 3) * 256 + (cw >> 24)];
+            float t0 = Tq[(m + 0) * 256 + (cw & 255)];
+            float t1 = Tq[(m + 1) * 256 + ((cw >> 8) & 255)];
+            float t2 = Tq[(m + 2) * 256 + ((cw >> 16) & 255)];
+            float t3 = Tq[(m + 3) * 256 + (cw >> 24)];
+            acc += (s0 + s1 + s2 + s3) - 2.0f * (t0 + t1 + t2 + t3);
           }
           acc -= 2.0f * dot;  // + qnorm at emit
         } else {
-          for (int32_t m = 0; m < M; m++) acc += Tq[m * 256 + code[m]];
+          for (int32_t m4 = 0; m4 < M / 4; m4++) {
+            const uint32_t cw = code4[m4];
+            const int32_t m = m4 * 4;
+            acc += Tq[(m + 0) * 256 + (cw & 255)] +
+                   Tq[(m + 1) * 256 + ((cw >> 8) & 255)] +
+                   Tq[(m + 2) * 256 + ((cw >> 16) & 255)] +
+                   Tq[(m + 3) * 256 + (cw >> 24)];
+          }
           acc = -(acc + dot);  // IP key = -score
         }
         const int64_t r = t0 + v;
