@@ -195,12 +195,20 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
   ix->events_ready = true;
   if (desc.kind == DG_INDEX_FLAT) ix->trained = true;
   if (desc.reserve > 0) {  // capacity hint: avoids grow-copies during add
-    if (dbuf_reserve(ix->d_vectors, (size_t)desc.reserve * desc.d * 4,
-                     ix->stream, false) != DG_OK ||
+    // PQ never stores raw vectors (codes only); others reserve the vector
+    // store up front
+    bool ok =
         dbuf_reserve(ix->d_ids, (size_t)desc.reserve * 8, ix->stream,
-                     false) != DG_OK ||
+                     false) == DG_OK &&
         dbuf_reserve(ix->d_assign, (size_t)desc.reserve * 4, ix->stream,
-                     false) != DG_OK) {
+                     false) == DG_OK;
+    if (ok && desc.kind == DG_INDEX_IVF_PQ)
+      ok = dbuf_reserve(ix->d_codes, (size_t)desc.reserve * desc.pq_m,
+                        ix->stream, false) == DG_OK;
+    if (ok && desc.kind != DG_INDEX_IVF_PQ)
+      ok = dbuf_reserve(ix->d_vectors, (size_t)desc.reserve * desc.d * 4,
+                        ix->stream, false) == DG_OK;
+    if (!ok) {
       dg_index_destroy(ix);
       return DG_ENOMEM;
     }
