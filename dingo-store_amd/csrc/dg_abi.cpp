@@ -8,6 +8,7 @@
 #include <cstdio>
 #include <cstring>
 #include <algorithm>
+#include <functional>
 #include <vector>
 
 #include "dg_internal.h"
@@ -953,10 +954,24 @@ static dg_status finalize_csr(dg_index* ix) {
 }
 
 // ---------------- search core ----------------
+// host-side monotone f32->u32 (mirror of enc_f32 in kernels.hip.cpp)
+static uint32_t h_enc_f32(float x) {
+  uint32_t u;
+  memcpy(&u, &x, 4);
+  return (int32_t)u < 0 ? ~u : (u | 0x80000000u);
+}
+
+struct dg_range_req {  // when set, search_core does radius search instead
+  float radius;        // faiss convention: L2 dist < r; IP score > r
+  int64_t* lims;       // caller's nq+1
+  int64_t** out_ids;   // malloc'd here
+  float** out_dists;
+};
+
 static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                              int32_t k, int32_t nprobe,
                              const dg_filter* filter, float* d_out_dist,
-                             int64_t* d_out_ids) {
+                             int64_t* d_out_ids, dg_range_req* rr = nullptr) {
   const int32_t d = ix->desc.d;
   const int metric = ix->desc.metric;
   const bool is_ivf = ix->desc.kind != DG_INDEX_FLAT;
@@ -1022,6 +1037,113 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     d_bitmap = (uint32_t*)ws_bitmap.p;
   }
 
+  // ---- range-search shared pieces ----
+  std::vector<uint64_t> h_thr;
+  dg_dbuf ws_range{};
+  int64_t* d_rcounts = nullptr;
+  uint64_t* d_thr = nullptr;
+  auto build_thr = [&]() -> dg_status {
+    h_thr.resize(nq);
+    if (metric == DG_METRIC_L2) {
+      std::vector<float> h_qn(nq);
+      DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+      DG_HIP_CHECK(hipMemcpy(h_qn.data(), dqn, (size_t)nq * 4,
+                             hipMemcpyDeviceToHost));
+      for (int64_t i = 0; i < nq; i++)
+        h_thr[i] = ((uint64_t)h_enc_f32(rr->radius - h_qn[i]) << 32);
+    } else {
+      for (int64_t i = 0; i < nq; i++)
+        h_thr[i] = ((uint64_t)h_enc_f32(-rr->radius) << 32);
+    }
+    dg_status s2 = dbuf_reserve(ws_range,
+                                (size_t)nq * 8 * 3 + ((size_t)nq + 1) * 8,
+                                ix->stream, false);
+    if (s2 != DG_OK) return s2;
+    d_thr = (uint64_t*)ws_range.p;
+    d_rcounts = (int64_t*)(d_thr + nq);
+    DG_HIP_CHECK(hipMemcpyAsync(d_thr, h_thr.data(), (size_t)nq * 8,
+                                hipMemcpyHostToDevice, ix->stream));
+    (void)hipMemsetAsync(d_rcounts, 0, (size_t)nq * 8, ix->stream);
+    return DG_OK;
+  };
+  // finish: lims on host -> compact via cb -> emit -> D2H + per-query sort
+  auto range_finish = [&](const std::function<dg_status(
+                              const int64_t* d_lims, int64_t* d_cursors,
+                              uint64_t* d_out)>& compact_cb) -> dg_status {
+    DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+    std::vector<int64_t> h_counts(nq);
+    DG_HIP_CHECK(hipMemcpy(h_counts.data(), d_rcounts, (size_t)nq * 8,
+                           hipMemcpyDeviceToHost));
+    rr->lims[0] = 0;
+    for (int64_t i = 0; i < nq; i++)
+      rr->lims[i + 1] = rr->lims[i] + h_counts[i];
+    const int64_t total = rr->lims[nq];
+    int64_t* d_lims = d_rcounts + nq;           // nq+1
+    int64_t* d_cursors = d_lims + nq + 1;       // reuse: need nq more
+    dg_status s2 = dbuf_reserve(ws_range,
+                                (size_t)nq * 8 * 3 + ((size_t)nq + 1) * 8,
+                                ix->stream, true);
+    if (s2 != DG_OK) return s2;
+    dg_dbuf d_res{};
+    if ((s2 = dbuf_reserve(d_res,
+                           (size_t)std::max<int64_t>(total, 1) * 20 + 16,
+                           ix->stream, false)) != DG_OK)
+      return s2;
+    uint64_t* d_packed = (uint64_t*)d_res.p;
+    float* d_dist = (float*)(d_packed + total);
+    int64_t* d_ids64 = (int64_t*)((char*)d_dist + ((total * 4 + 7) & ~7));
+    DG_HIP_CHECK(hipMemcpyAsync(d_lims, rr->lims, ((size_t)nq + 1) * 8,
+                                hipMemcpyHostToDevice, ix->stream));
+    (void)hipMemsetAsync(d_cursors, 0, (size_t)nq * 8, ix->stream);
+    s2 = compact_cb(d_lims, d_cursors, d_packed);
+    if (s2 != DG_OK) {
+      dbuf_free(d_res);
+      return s2;
+    }
+    dgk::range_emit(ix->stream, d_packed, d_lims,
+                    (const int64_t*)ix->d_csr_ids.p, dqn, nq, metric,
+                    metric == DG_METRIC_L2 ? 1 : 0, d_dist, d_ids64);
+    float* h_dist = (float*)malloc(std::max<int64_t>(total, 1) * 4);
+    int64_t* h_ids = (int64_t*)malloc(std::max<int64_t>(total, 1) * 8);
+    if (!h_dist || !h_ids) {
+      free(h_dist);
+      free(h_ids);
+      dbuf_free(d_res);
+      dg_set_error("host alloc failed");
+      return DG_ENOMEM;
+    }
+    DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+    if (total > 0) {
+      DG_HIP_CHECK(hipMemcpy(h_dist, d_dist, (size_t)total * 4,
+                             hipMemcpyDeviceToHost));
+      DG_HIP_CHECK(hipMemcpy(h_ids, d_ids64, (size_t)total * 8,
+                             hipMemcpyDeviceToHost));
+    }
+    dbuf_free(d_res);
+    // per-query sort best-first (L2 asc / IP desc), ties toward smaller id
+    for (int64_t qi = 0; qi < nq; qi++) {
+      int64_t a = rr->lims[qi], b = rr->lims[qi + 1];
+      std::vector<int64_t> ord(b - a);
+      for (int64_t i = 0; i < b - a; i++) ord[i] = a + i;
+      std::sort(ord.begin(), ord.end(), [&](int64_t x, int64_t y) {
+        float dx = h_dist[x], dy = h_dist[y];
+        if (dx != dy) return metric == DG_METRIC_L2 ? dx < dy : dx > dy;
+        return h_ids[x] < h_ids[y];
+      });
+      std::vector<float> td(b - a);
+      std::vector<int64_t> ti(b - a);
+      for (int64_t i = 0; i < b - a; i++) {
+        td[i] = h_dist[ord[i]];
+        ti[i] = h_ids[ord[i]];
+      }
+      memcpy(h_dist + a, td.data(), (size_t)(b - a) * 4);
+      memcpy(h_ids + a, ti.data(), (size_t)(b - a) * 8);
+    }
+    *rr->out_dists = h_dist;
+    *rr->out_ids = h_ids;
+    return DG_OK;
+  };
+
   if (!is_ivf) {
     // ---------- FLAT: chunked dots GEMM + dense select ----------
     const int64_t N = ix->ntotal;
@@ -1045,6 +1167,62 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     int64_t* bt = (int64_t*)(final_tk + (size_t)nq * k);   // base/total
     (void)hipEventRecord(ix->ev[1], ix->stream);  // no coarse stage in Flat
     (void)hipEventRecord(ix->ev[2], ix->stream);
+    const int mode = metric == DG_METRIC_L2 ? 1 : 2;
+    if (rr) {
+      // range search: count pass + compact pass (GEMM recomputed — range
+      // is not the measured hot path)
+      if ((st = build_thr()) != DG_OK) {
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        dbuf_free(ws_range);
+        return st;
+      }
+      for (int64_t ci = 0; ci < nchunks && st == DG_OK; ci++) {
+        int64_t c0 = ci * chunk_cols;
+        int64_t cc = std::min(chunk_cols, N - c0);
+        st = sgemm_dots(ix, dq, nq,
+                        (const float*)ix->d_csr_vectors.p + (size_t)c0 * d,
+                        cc, d, (float*)ix->ws_dots.p);
+        if (st != DG_OK) break;
+        dgk::count_below_dense(ix->stream, (const float*)ix->ws_dots.p,
+                               (const float*)ix->d_csr_vnorms.p + c0, nq, cc,
+                               mode, d_bitmap, c0, d_thr, d_rcounts);
+      }
+      if (st == DG_OK)
+        st = range_finish([&](const int64_t*, int64_t* d_cursors,
+                              uint64_t* d_out) -> dg_status {
+          dg_status s3 = DG_OK;
+          dg_dbuf d_off2{};
+          if ((s3 = dbuf_reserve(d_off2, ((size_t)nq + 1) * 8, ix->stream,
+                                 false)) != DG_OK)
+            return s3;
+          (void)hipMemcpyAsync(d_off2.p, rr->lims, ((size_t)nq + 1) * 8,
+                               hipMemcpyHostToDevice, ix->stream);
+          for (int64_t ci = 0; ci < nchunks && s3 == DG_OK; ci++) {
+            int64_t c0 = ci * chunk_cols;
+            int64_t cc = std::min(chunk_cols, N - c0);
+            s3 = sgemm_dots(ix, dq, nq,
+                            (const float*)ix->d_csr_vectors.p +
+                                (size_t)c0 * d,
+                            cc, d, (float*)ix->ws_dots.p);
+            if (s3 != DG_OK) break;
+            dgk::compact_below_dense(ix->stream,
+                                     (const float*)ix->ws_dots.p,
+                                     (const float*)ix->d_csr_vnorms.p + c0,
+                                     nq, cc, mode, d_bitmap, c0, d_thr,
+                                     (const int64_t*)d_off2.p, d_cursors,
+                                     d_out);
+          }
+          (void)hipStreamSynchronize(ix->stream);
+          dbuf_free(d_off2);
+          return s3;
+        });
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      dbuf_free(ws_range);
+      ix->times.last_nq = nq;
+      return st;
+    }
     for (int64_t ci = 0; ci < nchunks; ci++) {
       int64_t c0 = ci * chunk_cols;
       int64_t cc = std::min(chunk_cols, N - c0);
@@ -1054,8 +1232,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
       if (st != DG_OK) break;
       dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
                         (const float*)ix->d_csr_vnorms.p + c0, nq, cc, k,
-                        metric == DG_METRIC_L2 ? 1 : 2, d_bitmap, c0, slab,
-                        nchunks * k, ci * k);
+                        mode, d_bitmap, c0, slab, nchunks * k, ci * k);
     }
     (void)hipEventRecord(ix->ev[3], ix->stream);
     if (st == DG_OK) {
@@ -1264,6 +1441,33 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     }
     (void)hipEventRecord(ix->ev[3], ix->stream);
 
+    if (rr) {
+      // range search over the candidate buffer
+      if ((st = build_thr()) == DG_OK) {
+        dgk::count_below(ix->stream, (const uint64_t*)ix->ws_cand.p,
+                         q_cand_base, q_total, d_thr, nq, d_rcounts);
+        st = range_finish([&](const int64_t*, int64_t* d_cursors,
+                              uint64_t* d_out) -> dg_status {
+          dg_dbuf d_off2{};
+          dg_status s3 = dbuf_reserve(d_off2, ((size_t)nq + 1) * 8,
+                                      ix->stream, false);
+          if (s3 != DG_OK) return s3;
+          (void)hipMemcpyAsync(d_off2.p, rr->lims, ((size_t)nq + 1) * 8,
+                               hipMemcpyHostToDevice, ix->stream);
+          dgk::compact_below(ix->stream, (const uint64_t*)ix->ws_cand.p,
+                             q_cand_base, q_total, d_thr,
+                             (const int64_t*)d_off2.p, nq, d_cursors, d_out);
+          (void)hipStreamSynchronize(ix->stream);
+          dbuf_free(d_off2);
+          return DG_OK;
+        });
+      }
+      dbuf_free(d_fids);
+      dbuf_free(ws_bitmap);
+      dbuf_free(ws_range);
+      ix->times.last_nq = nq;
+      return st;
+    }
     // select + emit
     dgk::select_u64(ix->stream, (const uint64_t*)ix->ws_cand.p, q_cand_base,
                     q_total, nq, k, final_tk, k);
@@ -1357,13 +1561,49 @@ extern "C" dg_status dg_sync(dg_index* ix) {
   return DG_OK;
 }
 
-// ---------------- range search: §8f rank 2, not yet ----------------
-extern "C" dg_status dg_range_search(dg_index*, int64_t, const float*, float,
-                                     const dg_filter*, int64_t*, int64_t**,
-                                     float**) {
-  dg_set_error("range search lands with SURVEY.md §8f rank 2 "
-               "(EVECTOR_NOT_SUPPORT => reader brute-force fallback)");
-  return DG_ENOT_SUPPORT;
+// ---------------- range search ----------------
+// Restates VectorIndexIvfFlat::RangeSearch semantics
+// (src/vector/vector_index_ivf_flat.cc:278-368): radius in faiss
+// convention (L2: dist < radius; IP/cos: score > radius — the shim applies
+// the 1-r flip, ivf_flat.cc:302-305); CSR results, best-first per query.
+extern "C" dg_status dg_range_search(dg_index* ix, int64_t nq, const float* x,
+                                     float radius, const dg_filter* filter,
+                                     int64_t* lims, int64_t** out_ids,
+                                     float** out_dists) {
+  if (!ix || !x || !lims || !out_ids || !out_dists || nq <= 0) {
+    dg_set_error("bad range search args");
+    return DG_EINVAL;
+  }
+  DeviceGuard g(ix->device);
+  if (!ix->trained || ix->ntotal == 0 ||
+      (ix->desc.kind == DG_INDEX_IVF_PQ && !ix->pq_trained)) {
+    memset(lims, 0, ((size_t)nq + 1) * 8);
+    *out_ids = (int64_t*)malloc(8);
+    *out_dists = (float*)malloc(4);
+    return DG_OK;
+  }
+  {
+    std::unique_lock lk(ix->rw, std::defer_lock);
+    if (!ix->csr_valid) {
+      lk.lock();
+      if (!ix->csr_valid) {
+        dg_status st = finalize_csr(ix);
+        if (st != DG_OK) return st;
+      }
+    }
+  }
+  std::shared_lock lk(ix->rw);
+  static thread_local dg_dbuf t_in;
+  dg_status st;
+  if ((st = dbuf_reserve(t_in, (size_t)nq * ix->desc.d * 4, ix->stream,
+                         false)) != DG_OK)
+    return st;
+  DG_HIP_CHECK(hipMemcpyAsync(t_in.p, x, (size_t)nq * ix->desc.d * 4,
+                              hipMemcpyHostToDevice, ix->stream));
+  dg_range_req rr{radius, lims, out_ids, out_dists};
+  // nprobe: reference RangeSearch uses the index default (clamped); pass 0
+  return search_core(ix, nq, (const float*)t_in.p, 1, 0, filter, nullptr,
+                     nullptr, &rr);
 }
 
 extern "C" void dg_free(void* p) { free(p); }

@@ -199,6 +199,27 @@ void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
                       const int64_t* csr_offsets, const int32_t* chunk_off,
                       const int64_t* chunk_base, const float* rowmajor,
                       int32_t d, int32_t chunk_rows, float* tvec);
+// range search
+void range_emit(hipStream_t s, const uint64_t* packed, const int64_t* lims,
+                const int64_t* ids_lookup, const float* qnorms, int64_t nq,
+                int metric, int add_qnorm, float* out_dist,
+                int64_t* out_ids);
+void count_below(hipStream_t s, const uint64_t* cand, const int64_t* base,
+                 const int64_t* total, const uint64_t* thr, int64_t nq,
+                 int64_t* counts);
+void compact_below(hipStream_t s, const uint64_t* cand, const int64_t* base,
+                   const int64_t* total, const uint64_t* thr,
+                   const int64_t* out_off, int64_t nq, int64_t* cursors,
+                   uint64_t* out);
+void count_below_dense(hipStream_t s, const float* scores,
+                       const float* cnorms, int64_t rows, int64_t cols,
+                       int mode, const uint32_t* bitmap, int64_t col_base,
+                       const uint64_t* thr, int64_t* counts);
+void compact_below_dense(hipStream_t s, const float* scores,
+                         const float* cnorms, int64_t rows, int64_t cols,
+                         int mode, const uint32_t* bitmap, int64_t col_base,
+                         const uint64_t* thr, const int64_t* out_off,
+                         int64_t* cursors, uint64_t* out);
 // IVF-PQ
 void residual(hipStream_t s, const float* x, const int32_t* assign,
               const float* centroids, int64_t n, int32_t d, float* out);

@@ -466,6 +466,133 @@ __global__ void __launch_bounds__(256, 2) k_ivf_scan(
   }
 }
 
+// ---------- range search (radius) ----------
+// candidates (packed u64) below a per-query threshold key.  Strict '<'
+// matches faiss RangeSearch (L2: dist < radius; IP: score > radius).
+__global__ void k_count_below(const uint64_t* __restrict__ cand,
+                              const int64_t* __restrict__ base,
+                              const int64_t* __restrict__ total,
+                              const uint64_t* __restrict__ thr, int64_t nq,
+                              int64_t* __restrict__ counts) {
+  __shared__ int64_t lds[256];
+  int64_t q = blockIdx.x;
+  if (q >= nq) return;
+  const uint64_t* seg = cand + base[q];
+  const uint64_t t = thr[q];
+  int64_t c = 0;
+  for (int64_t i = threadIdx.x; i < total[q]; i += blockDim.x)
+    if (seg[i] < t) c++;
+  lds[threadIdx.x] = c;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s; s >>= 1) {
+    if (threadIdx.x < s) lds[threadIdx.x] += lds[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) counts[q] = lds[0];
+}
+
+__global__ void k_compact_below(const uint64_t* __restrict__ cand,
+                                const int64_t* __restrict__ base,
+                                const int64_t* __restrict__ total,
+                                const uint64_t* __restrict__ thr,
+                                const int64_t* __restrict__ out_off,
+                                int64_t nq, int64_t* __restrict__ cursors,
+                                uint64_t* __restrict__ out) {
+  int64_t q = blockIdx.x;
+  if (q >= nq) return;
+  const uint64_t* seg = cand + base[q];
+  const uint64_t t = thr[q];
+  for (int64_t i = threadIdx.x; i < total[q]; i += blockDim.x) {
+    uint64_t c = seg[i];
+    if (c < t) {
+      int64_t pos = atomicAdd((unsigned long long*)&cursors[q], 1ull);
+      out[out_off[q] + pos] = c;  // host sorts per query afterwards
+    }
+  }
+}
+
+// dense (Flat) variants over a dots chunk; mode as k_select_dense
+__global__ void k_count_below_dense(const float* __restrict__ scores,
+                                    const float* __restrict__ cnorms,
+                                    int64_t rows, int64_t cols, int mode,
+                                    const uint32_t* __restrict__ bitmap,
+                                    int64_t col_base,
+                                    const uint64_t* __restrict__ thr,
+                                    int64_t* __restrict__ counts) {
+  __shared__ int64_t lds[256];
+  int64_t row = blockIdx.x;
+  if (row >= rows) return;
+  const float* sr = scores + row * cols;
+  const uint64_t t = thr[row];
+  int64_t c = 0;
+  for (int64_t j = threadIdx.x; j < cols; j += blockDim.x) {
+    int64_t g = col_base + j;
+    if (bitmap && !((bitmap[g >> 5] >> (g & 31)) & 1)) continue;
+    float s = sr[j];
+    float key = (mode == 0) ? s : (mode == 1) ? cnorms[j] - 2.0f * s : -s;
+    if (pack_cand(key, (uint32_t)g) < t) c++;
+  }
+  lds[threadIdx.x] = c;
+  __syncthreads();
+  for (int s2 = blockDim.x / 2; s2; s2 >>= 1) {
+    if (threadIdx.x < s2) lds[threadIdx.x] += lds[threadIdx.x + s2];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) counts[row] += lds[0];  // accumulates over chunks
+}
+
+__global__ void k_compact_below_dense(const float* __restrict__ scores,
+                                      const float* __restrict__ cnorms,
+                                      int64_t rows, int64_t cols, int mode,
+                                      const uint32_t* __restrict__ bitmap,
+                                      int64_t col_base,
+                                      const uint64_t* __restrict__ thr,
+                                      const int64_t* __restrict__ out_off,
+                                      int64_t* __restrict__ cursors,
+                                      uint64_t* __restrict__ out) {
+  int64_t row = blockIdx.x;
+  if (row >= rows) return;
+  const float* sr = scores + row * cols;
+  const uint64_t t = thr[row];
+  for (int64_t j = threadIdx.x; j < cols; j += blockDim.x) {
+    int64_t g = col_base + j;
+    if (bitmap && !((bitmap[g >> 5] >> (g & 31)) & 1)) continue;
+    float s = sr[j];
+    float key = (mode == 0) ? s : (mode == 1) ? cnorms[j] - 2.0f * s : -s;
+    uint64_t c = pack_cand(key, (uint32_t)g);
+    if (c < t) {
+      int64_t pos = atomicAdd((unsigned long long*)&cursors[row], 1ull);
+      out[out_off[row] + pos] = c;
+    }
+  }
+}
+
+__global__ void k_range_emit(const uint64_t* __restrict__ packed,
+                             const int64_t* __restrict__ lims,
+                             const int64_t* __restrict__ ids_lookup,
+                             const float* __restrict__ qnorms, int64_t nq,
+                             int metric, int add_qnorm,
+                             float* __restrict__ out_dist,
+                             int64_t* __restrict__ out_ids) {
+  int64_t q = blockIdx.x;
+  if (q >= nq) return;
+  for (int64_t i = lims[q] + threadIdx.x; i < lims[q + 1];
+       i += blockDim.x) {
+    uint64_t c = packed[i];
+    float key = dec_f32((uint32_t)(c >> 32));
+    uint32_t row = (uint32_t)c;
+    float dist;
+    if (metric == 0) {
+      dist = add_qnorm ? key + qnorms[q] : key;
+      if (dist < 0.f) dist = 0.f;
+    } else {
+      dist = -key;
+    }
+    out_dist[i] = dist;
+    out_ids[i] = ids_lookup ? ids_lookup[row] : (int64_t)row;
+  }
+}
+
 // ---------- misc small kernels ----------
 __global__ void k_gather_rows_by_index(const float* __restrict__ src,
                                        const int64_t* __restrict__ idx,
@@ -1146,6 +1273,49 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     default: DG_SCAN_LAUNCH(16, 4); break;
   }
 #undef DG_SCAN_LAUNCH
+}
+
+void range_emit(hipStream_t s, const uint64_t* packed, const int64_t* lims,
+                const int64_t* ids_lookup, const float* qnorms, int64_t nq,
+                int metric, int add_qnorm, float* out_dist,
+                int64_t* out_ids) {
+  hipLaunchKernelGGL(k_range_emit, dim3((uint32_t)nq), dim3(256), 0, s,
+                     packed, lims, ids_lookup, qnorms, nq, metric, add_qnorm,
+                     out_dist, out_ids);
+}
+
+void count_below(hipStream_t s, const uint64_t* cand, const int64_t* base,
+                 const int64_t* total, const uint64_t* thr, int64_t nq,
+                 int64_t* counts) {
+  hipLaunchKernelGGL(k_count_below, dim3((uint32_t)nq), dim3(256), 0, s,
+                     cand, base, total, thr, nq, counts);
+}
+
+void compact_below(hipStream_t s, const uint64_t* cand, const int64_t* base,
+                   const int64_t* total, const uint64_t* thr,
+                   const int64_t* out_off, int64_t nq, int64_t* cursors,
+                   uint64_t* out) {
+  hipLaunchKernelGGL(k_compact_below, dim3((uint32_t)nq), dim3(256), 0, s,
+                     cand, base, total, thr, out_off, nq, cursors, out);
+}
+
+void count_below_dense(hipStream_t s, const float* scores,
+                       const float* cnorms, int64_t rows, int64_t cols,
+                       int mode, const uint32_t* bitmap, int64_t col_base,
+                       const uint64_t* thr, int64_t* counts) {
+  hipLaunchKernelGGL(k_count_below_dense, dim3((uint32_t)rows), dim3(256), 0,
+                     s, scores, cnorms, rows, cols, mode, bitmap, col_base,
+                     thr, counts);
+}
+
+void compact_below_dense(hipStream_t s, const float* scores,
+                         const float* cnorms, int64_t rows, int64_t cols,
+                         int mode, const uint32_t* bitmap, int64_t col_base,
+                         const uint64_t* thr, const int64_t* out_off,
+                         int64_t* cursors, uint64_t* out) {
+  hipLaunchKernelGGL(k_compact_below_dense, dim3((uint32_t)rows), dim3(256),
+                     0, s, scores, cnorms, rows, cols, mode, bitmap,
+                     col_base, thr, out_off, cursors, out);
 }
 
 void residual(hipStream_t s, const float* x, const int32_t* assign,

@@ -250,6 +250,33 @@ class Index:
                                       C.c_void_p(ids_ptr)),
                "dg_search_device")
 
+    def range_search(self, queries, radius, filt=None):
+        """Radius search (faiss convention: L2 dist < r; IP score > r).
+        Returns (lims[nq+1], dists, ids) with per-query best-first order."""
+        q = np.ascontiguousarray(queries, np.float32)
+        nq = q.shape[0]
+        lims = np.zeros(nq + 1, np.int64)
+        out_ids = C.POINTER(C.c_int64)()
+        out_dists = C.POINTER(C.c_float)()
+        l = lib()
+        l.dg_range_search.argtypes = [
+            C.c_void_p, C.c_int64, _f32p, C.c_float, C.POINTER(_Filter),
+            _i64p, C.POINTER(C.POINTER(C.c_int64)),
+            C.POINTER(C.POINTER(C.c_float)),
+        ]
+        l.dg_free.argtypes = [C.c_void_p]
+        fp = C.byref(filt) if filt is not None else None
+        _check(l.dg_range_search(self.h, nq, q, radius, fp, lims,
+                                 C.byref(out_ids), C.byref(out_dists)),
+               "dg_range_search")
+        total = int(lims[-1])
+        dists = np.ctypeslib.as_array(out_dists, (max(total, 1),))[
+            :total].copy()
+        ids = np.ctypeslib.as_array(out_ids, (max(total, 1),))[:total].copy()
+        l.dg_free(C.cast(out_ids, C.c_void_p))
+        l.dg_free(C.cast(out_dists, C.c_void_p))
+        return lims, dists, ids
+
     def sync(self):
         _check(lib().dg_sync(self.h), "dg_sync")
 

@@ -189,11 +189,48 @@ class GpuIndexBase : public VectorIndex {
     return Status::OK();
   }
 
-  Status RangeSearch(const std::vector<VectorWithId>&, float,
-                     const std::vector<std::shared_ptr<FilterFunctor>>&, bool,
-                     const VectorSearchParameter&,
-                     std::vector<VectorWithDistanceResult>&) override {
-    return {kEVectorNotSupport, "range search: SURVEY.md 8f rank 2"};
+  Status RangeSearch(const std::vector<VectorWithId>& queries, float radius,
+                     const std::vector<std::shared_ptr<FilterFunctor>>& fs,
+                     bool, const VectorSearchParameter&,
+                     std::vector<VectorWithDistanceResult>& results) override {
+    // mirrors VectorIndexIvfFlat::RangeSearch: for IP/cosine the dingo
+    // radius converts to a faiss score threshold 1 - r
+    // (src/vector/vector_index_ivf_flat.cc:302-305)
+    if (queries.empty())
+      return {kEillegalParamteters, "vector_with_ids is empty"};
+    std::vector<float> x;
+    std::vector<int64_t> ids;
+    Status s = pack(queries, dim_, x, ids);
+    if (!s.ok()) return s;
+    dg_filter df{};
+    dg_filter* dfp = nullptr;
+    if (!fs.empty()) {
+      if (fs.size() == 1 && fs[0]->ToDeviceFilter(&df)) dfp = &df;
+      else return {kEVectorNotSupport, "composite filters via reader"};
+    }
+    float r = metric_ == MetricType::kL2 ? radius : 1.0f - radius;
+    std::vector<int64_t> lims(queries.size() + 1);
+    int64_t* rids = nullptr;
+    float* rdists = nullptr;
+    dg_status st = dg_range_search(idx_, (int64_t)queries.size(), x.data(),
+                                   r, dfp, lims.data(), &rids, &rdists);
+    if (st != DG_OK) return from_dg(st);
+    results.clear();
+    results.resize(queries.size());
+    for (size_t q = 0; q < queries.size(); q++) {
+      for (int64_t i = lims[q]; i < lims[q + 1]; i++) {
+        VectorWithDistance vd;
+        vd.vector_with_id.id = rids[i];
+        vd.vector_with_id.vector.dimension = dim_;
+        vd.metric_type = metric_;
+        vd.distance = (metric_ == MetricType::kL2) ? rdists[i]
+                                                   : 1.0f - rdists[i];
+        results[q].vector_with_distances.push_back(std::move(vd));
+      }
+    }
+    dg_free(rids);
+    dg_free(rdists);
+    return Status::OK();
   }
 
  protected:

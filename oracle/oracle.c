@@ -538,3 +538,89 @@ void dgo_ivfpq_search(int metric, int32_t nlist, int32_t d,
     free(res);
   }
 }
+
+/* ---------------- range search ----------------
+ * faiss RangeSearch semantics restated (reference RangeSearch call sites,
+ * vector_index_flat.cc:268+ / vector_index_ivf_flat.cc:278-368): L2 keeps
+ * dist < radius, IP/cos keeps score > radius (the 1-r flip happens in the
+ * shim).  Results per query sorted best-first, ties toward smaller id.
+ * Caller provides capacity; results beyond cap are counted in lims but
+ * dropped (tests size cap generously). */
+static void range_collect(int metric, const float* qv, int32_t d,
+                          const float* vecs, const int64_t* ids_arr,
+                          int64_t lo, int64_t hi, float radius,
+                          dgo_cand** buf, int64_t* cnt, int64_t* cap_now) {
+  float thr = (metric == DGO_L2) ? radius : -radius;
+  for (int64_t i = lo; i < hi; i++) {
+    float key = metric_key(metric, qv, vecs + (size_t)i * d, d);
+    if (key < thr) {
+      if (*cnt == *cap_now) {
+        *cap_now = *cap_now * 2 + 64;
+        *buf = (dgo_cand*)realloc(*buf, sizeof(dgo_cand) * *cap_now);
+      }
+      (*buf)[*cnt].key = key;
+      (*buf)[*cnt].id = ids_arr ? ids_arr[i] : i;
+      (*cnt)++;
+    }
+  }
+}
+
+void dgo_flat_range_search(int metric, int64_t n, int32_t d,
+                           const float* base, const int64_t* ids, int64_t nq,
+                           const float* queries, float radius, int64_t* lims,
+                           int64_t cap, float* out_dist, int64_t* out_ids) {
+  lims[0] = 0;
+  for (int64_t q = 0; q < nq; q++) {
+    dgo_cand* buf = NULL;
+    int64_t cnt = 0, cap_now = 0;
+    range_collect(metric, queries + (size_t)q * d, d, base, ids, 0, n,
+                  radius, &buf, &cnt, &cap_now);
+    qsort(buf, cnt, sizeof(dgo_cand), cand_cmp);
+    int64_t base_off = lims[q];
+    for (int64_t i = 0; i < cnt && base_off + i < cap; i++) {
+      out_dist[base_off + i] =
+          (metric == DGO_L2) ? buf[i].key : -buf[i].key;
+      out_ids[base_off + i] = buf[i].id;
+    }
+    lims[q + 1] = base_off + cnt;
+    free(buf);
+  }
+}
+
+void dgo_ivf_range_search(int metric, int32_t nlist, int32_t d,
+                          const float* centroids, const int64_t* offsets,
+                          const float* grouped_vectors,
+                          const int64_t* grouped_ids, int64_t nq,
+                          const float* queries, float radius, int32_t nprobe,
+                          int64_t* lims, int64_t cap, float* out_dist,
+                          int64_t* out_ids) {
+  if (nprobe > nlist) nprobe = nlist;
+  lims[0] = 0;
+  dgo_cand* pstorage = (dgo_cand*)malloc(sizeof(dgo_cand) * nprobe);
+  for (int64_t q = 0; q < nq; q++) {
+    const float* qv = queries + (size_t)q * d;
+    dgo_topk probes;
+    topk_init(&probes, pstorage, nprobe);
+    for (int32_t l = 0; l < nlist; l++)
+      topk_push(&probes, metric_key(metric, qv, centroids + (size_t)l * d, d),
+                l);
+    qsort(probes.c, probes.size, sizeof(dgo_cand), cand_cmp);
+    dgo_cand* buf = NULL;
+    int64_t cnt = 0, cap_now = 0;
+    for (int32_t p = 0; p < probes.size; p++) {
+      int32_t l = (int32_t)probes.c[p].id;
+      range_collect(metric, qv, d, grouped_vectors, grouped_ids, offsets[l],
+                    offsets[l + 1], radius, &buf, &cnt, &cap_now);
+    }
+    qsort(buf, cnt, sizeof(dgo_cand), cand_cmp);
+    int64_t base_off = lims[q];
+    for (int64_t i = 0; i < cnt && base_off + i < cap; i++) {
+      out_dist[base_off + i] =
+          (metric == DGO_L2) ? buf[i].key : -buf[i].key;
+      out_ids[base_off + i] = buf[i].id;
+    }
+    lims[q + 1] = base_off + cnt;
+    free(buf);
+  }
+  free(pstorage);
+}

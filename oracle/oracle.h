@@ -130,6 +130,21 @@ void dgo_ivf_search_fast(int metric, int32_t nlist, int32_t d,
                          const uint8_t* list_mask, float* out_dist,
                          int64_t* out_ids);
 
+
+/* --- range search (faiss RangeSearch restated; results best-first, ties
+ * toward smaller id; lims always filled, entries beyond cap dropped) --- */
+void dgo_flat_range_search(int metric, int64_t n, int32_t d,
+                           const float* base, const int64_t* ids, int64_t nq,
+                           const float* queries, float radius, int64_t* lims,
+                           int64_t cap, float* out_dist, int64_t* out_ids);
+void dgo_ivf_range_search(int metric, int32_t nlist, int32_t d,
+                          const float* centroids, const int64_t* offsets,
+                          const float* grouped_vectors,
+                          const int64_t* grouped_ids, int64_t nq,
+                          const float* queries, float radius, int32_t nprobe,
+                          int64_t* lims, int64_t cap, float* out_dist,
+                          int64_t* out_ids);
+
 const char* dgo_version(void);
 
 #ifdef __cplusplus

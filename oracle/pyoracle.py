@@ -179,6 +179,45 @@ def ivf_search_indexed_fast(metric, centroids, member_offsets, member_rows,
     return out_dist, out_ids
 
 
+def flat_range_search(metric, base, queries, radius, ids=None, cap=1 << 22):
+    base = np.ascontiguousarray(base, np.float32)
+    queries = np.ascontiguousarray(queries, np.float32)
+    n, d = base.shape
+    nq = queries.shape[0]
+    lims = np.zeros(nq + 1, np.int64)
+    out_dist = np.empty(cap, np.float32)
+    out_ids = np.empty(cap, np.int64)
+    fn = _lib.dgo_flat_range_search
+    idp = (np.ascontiguousarray(ids, np.int64).ctypes.data_as(C.c_void_p)
+           if ids is not None else None)
+    fn.argtypes = [C.c_int, C.c_int64, C.c_int32, _f32p, C.c_void_p,
+                   C.c_int64, _f32p, C.c_float, _i64p, C.c_int64, _f32p,
+                   _i64p]
+    fn(metric, n, d, base, idp, nq, queries, radius, lims, cap, out_dist,
+       out_ids)
+    t = int(lims[-1])
+    return lims, out_dist[:t].copy(), out_ids[:t].copy()
+
+
+def ivf_range_search(metric, centroids, offsets, gv, gi, queries, radius,
+                     nprobe, cap=1 << 22):
+    centroids = np.ascontiguousarray(centroids, np.float32)
+    queries = np.ascontiguousarray(queries, np.float32)
+    nlist, d = centroids.shape
+    nq = queries.shape[0]
+    lims = np.zeros(nq + 1, np.int64)
+    out_dist = np.empty(cap, np.float32)
+    out_ids = np.empty(cap, np.int64)
+    fn = _lib.dgo_ivf_range_search
+    fn.argtypes = [C.c_int, C.c_int32, C.c_int32, _f32p, _i64p, _f32p,
+                   _i64p, C.c_int64, _f32p, C.c_float, C.c_int32, _i64p,
+                   C.c_int64, _f32p, _i64p]
+    fn(metric, nlist, d, centroids, offsets, gv, gi, nq, queries, radius,
+       nprobe, lims, cap, out_dist, out_ids)
+    t = int(lims[-1])
+    return lims, out_dist[:t].copy(), out_ids[:t].copy()
+
+
 def pq_train(residuals, m, nbits=8, seed=1234):
     residuals = np.ascontiguousarray(residuals, np.float32)
     n, d = residuals.shape

@@ -469,3 +469,73 @@ def test_ivfpq_untrained_and_edge():
                                                            np.float32))
     finally:
         idx.close()
+
+
+# ---------------- range search (§8f rank 2) ----------------
+def test_range_search_flat():
+    base, q = make_data(n=10000, d=64, nq=32)
+    idx = dg.Index(dg.FLAT, dg.L2, 64)
+    try:
+        idx.add(np.arange(10000, dtype=np.int64), base)
+        # radius around the typical NN distance so results are non-trivial
+        gd, gi = idx.search(q, 10)
+        radius = float(np.median(gd[:, 5]))
+        lims, dists, ids = idx.range_search(q, radius)
+    finally:
+        idx.close()
+    ol, od, oi = orc.flat_range_search(orc.L2, base, q, radius)
+    assert np.array_equal(lims, ol)
+    assert np.array_equal(ids, oi)
+    assert np.allclose(dists, od, rtol=1e-3, atol=1e-3)
+    assert (dists < radius).all()
+
+
+def test_range_search_ivf():
+    base, q = make_data(n=20000, d=64, nq=32)
+    nlist, nprobe = 32, 32  # all lists => same set as flat
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        gd, _ = gpu.search(q, 10, nprobe=nprobe)
+        radius = float(np.median(gd[:, 5]))
+        lims, dists, ids = gpu.range_search(q, radius)
+    finally:
+        gpu.close()
+    ol, od, oi = orc.ivf_range_search(orc.L2, cents, off, gv, gi_, q, radius,
+                                      nprobe)
+    assert np.array_equal(lims, ol)
+    assert np.array_equal(ids, oi)
+    assert np.allclose(dists, od, rtol=1e-3, atol=1e-3)
+
+
+def test_range_search_ip_and_filter():
+    base, q = make_data(n=5000, d=32, nq=8)
+    idx = dg.Index(dg.FLAT, dg.IP, 32)
+    try:
+        idx.add(np.arange(5000, dtype=np.int64), base)
+        gd, _ = idx.search(q, 10)
+        radius = float(np.median(gd[:, 5]))  # raw score threshold
+        lims, dists, ids = idx.range_search(q, radius)
+        assert (dists > radius).all()
+        f = dg.make_filter(kind=1, min_id=0, max_id=1000)
+        lf, df_, if_ = idx.range_search(q, radius, filt=f)
+        assert (if_ < 1000).all()
+    finally:
+        idx.close()
+
+
+# ---------------- concurrent searches (thread-safety contract) ----------
+def test_concurrent_search():
+    from concurrent.futures import ThreadPoolExecutor
+    base, q = make_data(n=20000, d=64, nq=16)
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, 32)
+    try:
+        ref_d, ref_i = gpu.search(q, 10, nprobe=8)
+
+        def worker(i):
+            d_, i_ = gpu.search(q, 10, nprobe=8)
+            return np.array_equal(i_, ref_i)
+
+        with ThreadPoolExecutor(8) as ex:
+            assert all(ex.map(worker, range(16)))
+    finally:
+        gpu.close()
