@@ -59,9 +59,13 @@ static dg_status dbuf_reserve(dg_dbuf& b, size_t bytes, hipStream_t s,
       dg_set_error("grow copy failed");
       return DG_EINTERNAL;
     }
-    (void)hipStreamSynchronize(s);
   }
-  if (b.p) (void)hipFree(b.p);
+  if (b.p) {
+    // drain pending work before freeing: earlier enqueued kernels may still
+    // reference the old allocation (growth is rare; steady state untouched)
+    (void)hipStreamSynchronize(s);
+    (void)hipFree(b.p);
+  }
   b.p = np;
   b.cap = newcap;
   b.bytes = bytes;
@@ -1518,6 +1522,7 @@ extern "C" dg_status dg_search_device(dg_index* ix, int64_t nq,
     }
   }
   std::shared_lock lk(ix->rw);
+  std::lock_guard sg(ix->search_mu);  // workspaces + stream are shared
   return search_core(ix, nq, d_x, k, nprobe, filter, d_out_dist, d_out_ids);
 }
 
@@ -1602,6 +1607,7 @@ extern "C" dg_status dg_range_search(dg_index* ix, int64_t nq, const float* x,
                               hipMemcpyHostToDevice, ix->stream));
   dg_range_req rr{radius, lims, out_ids, out_dists};
   // nprobe: reference RangeSearch uses the index default (clamped); pass 0
+  std::lock_guard sg(ix->search_mu);
   return search_core(ix, nq, (const float*)t_in.p, 1, 0, filter, nullptr,
                      nullptr, &rr);
 }

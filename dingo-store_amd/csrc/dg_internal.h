@@ -127,6 +127,11 @@ struct dg_index {
   bool events_ready = false;
 
   std::shared_mutex rw;  // search shared; mutation exclusive
+  // Concurrent dg_search calls are SAFE but serialized per index: searches
+  // share the workspace buffers and the HIP stream, so execution holds this
+  // mutex (the reference serializes device work on one stream anyway;
+  // intra-index parallelism comes from batching, SURVEY.md §8b threading).
+  std::mutex search_mu;
 };
 
 // ---- kernel launchers (kernels.hip.cpp; authoritative signatures) ----
