@@ -1270,6 +1270,7 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     case 2: DG_SCAN_LAUNCH(16, 2); break;
     case 3: DG_SCAN_LAUNCH(8, 8); break;
     case 4: DG_SCAN_LAUNCH(4, 4); break;
+    case 5: DG_SCAN_LAUNCH(12, 4); break;
     default: DG_SCAN_LAUNCH(16, 4); break;
   }
 #undef DG_SCAN_LAUNCH
