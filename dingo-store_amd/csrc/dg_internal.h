@@ -163,8 +163,6 @@ void argmin_rows(hipStream_t s, const float* dots, const float* cnorms,
 void build_pass_bitmap(hipStream_t s, const int64_t* ids, int64_t n,
                        const dg_dev_filter* f, uint32_t* bitmap);
 // IVF probe machinery
-void probe_mask_apply(hipStream_t s, int32_t* probes, int64_t nq,
-                      int32_t nprobe, const uint8_t* mask);
 void hist_probes(hipStream_t s, const int32_t* probes, int64_t nq,
                  int32_t nprobe, int32_t nlist, int32_t* counts);
 void scatter_probes(hipStream_t s, const int32_t* probes, int64_t nq,
@@ -182,16 +180,6 @@ void fill_units(hipStream_t s, const int32_t* unit_offsets,
                 const int64_t* csr_offsets, int32_t chunk_rows,
                 uint32_t* units /* 2 x u32 per unit: list, chunk */,
                 int32_t total_units);
-// THE dominant kernel: grouped inverted-list scan.
-void ivf_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
-              const int64_t* csr_offsets, const float* csr_vectors,
-              const float* csr_vnorms, const int64_t* csr_ids,
-              const float* queries, const float* qnorms, int32_t d,
-              const int32_t* inv_offsets, const int32_t* inv_q,
-              const int32_t* inv_rank, const int64_t* qp_off,
-              const int64_t* q_cand_base, int metric,
-              const uint32_t* pass_bitmap, int32_t chunk_rows, int32_t nprobe,
-              int32_t qt_max, uint64_t* cand);
 void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                   const int64_t* csr_offsets, const int32_t* chunk_off,
                   const int64_t* chunk_base, const float* tvec,
@@ -262,8 +250,6 @@ void cluster_means(hipStream_t s, const float* grouped, const int64_t* offsets,
 void iota_i32(hipStream_t s, int32_t* p, int64_t n, int32_t value);
 void excl_scan_i32_to_i64(hipStream_t s, const int32_t* in, int32_t n,
                           int64_t* out /* n+1 */);
-void excl_scan_i32(hipStream_t s, const int32_t* in, int32_t n,
-                   int32_t* out /* n+1 */);
 void excl_scan_i64(hipStream_t s, const int64_t* in, int64_t n,
                    int64_t* out /* n+1 */);
 }  // namespace dgk

@@ -357,115 +357,6 @@ __global__ void k_fill_units(const int32_t* __restrict__ unit_offsets,
   }
 }
 
-// ---------- THE dominant kernel: grouped inverted-list scan ----------
-// One block per (list, chunk).  The chunk's rows are read ONCE; all queries
-// probing the list are processed in LDS-staged tiles of QT.  Wave-per-row,
-// float4 coalesced; per (row, query) distance by wave shuffle reduction;
-// candidates written as packed u64 at precomputed dense offsets.
-// PER_LANE (= ceil(d/256)) is a template parameter so the row registers stay
-// in VGPRs (a runtime bound would spill the array to scratch) and the next
-// row is prefetched while the current one is reduced (2 rows in flight per
-// wave on top of wave-level parallelism).
-template <int PER_LANE>
-__global__ void __launch_bounds__(256, 2) k_ivf_scan(
-    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
-    const float* __restrict__ csr_vectors, const float* __restrict__ csr_vnorms,
-    const float* __restrict__ queries, int32_t d,
-    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
-    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
-    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
-    const uint32_t* __restrict__ bitmap, int32_t chunk_rows, int32_t qt_max,
-    uint64_t* __restrict__ cand) {
-  extern __shared__ float smem[];          // [qt_max * d] query tile
-  int64_t* cbase = (int64_t*)(smem + (size_t)qt_max * d);  // [qt_max]
-
-  const uint32_t list = units[2 * blockIdx.x];
-  const uint32_t chunk = units[2 * blockIdx.x + 1];
-  const int64_t list_start = csr_offsets[list];
-  const int64_t list_end = csr_offsets[list + 1];
-  const int64_t row_start = list_start + (int64_t)chunk * chunk_rows;
-  const int64_t row_end = min(list_end, row_start + chunk_rows);
-  const int32_t iq0 = inv_offsets[list];
-  const int32_t nql = inv_offsets[list + 1] - iq0;
-
-  const int wave_id = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  const int nwaves = blockDim.x / WAVE;
-  const int d4 = d / 4;  // d must be a multiple of 4 (checked host-side)
-
-  for (int32_t t0 = 0; t0 < nql; t0 += qt_max) {
-    const int32_t qt = min(qt_max, nql - t0);
-    // stage qt query vectors + their candidate bases
-    __syncthreads();
-    for (int32_t j = 0; j < qt; j++) {
-      int32_t q = inv_q[iq0 + t0 + j];
-      const float4* src = (const float4*)(queries + (size_t)q * d);
-      float4* dst = (float4*)(smem + (size_t)j * d);
-      for (int i = threadIdx.x; i < d4; i += blockDim.x) dst[i] = src[i];
-    }
-    if (threadIdx.x < qt) {
-      int32_t j = threadIdx.x;
-      int32_t q = inv_q[iq0 + t0 + j];
-      int32_t rank = inv_rank[iq0 + t0 + j];
-      // cand position for row r = q's base + this probe's offset +
-      // (r - list_start)
-      cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
-                 list_start;
-    }
-    __syncthreads();
-    const int64_t my_cbase = (lane < qt) ? cbase[lane] : 0;
-
-    float4 cur[PER_LANE], nxt[PER_LANE];
-    int64_t r = row_start + wave_id;
-    if (r < row_end) {
-      const float4* v4 = (const float4*)(csr_vectors + (size_t)r * d);
-#pragma unroll
-      for (int i = 0; i < PER_LANE; i++) {
-        int idx = lane + i * WAVE;
-        cur[i] = (idx < d4) ? v4[idx] : make_float4(0.f, 0.f, 0.f, 0.f);
-      }
-    }
-    for (; r < row_end; r += nwaves) {
-      const int64_t rn = r + nwaves;
-      if (rn < row_end) {  // prefetch next row while reducing this one
-        const float4* v4n = (const float4*)(csr_vectors + (size_t)rn * d);
-#pragma unroll
-        for (int i = 0; i < PER_LANE; i++) {
-          int idx = lane + i * WAVE;
-          nxt[i] = (idx < d4) ? v4n[idx] : make_float4(0.f, 0.f, 0.f, 0.f);
-        }
-      }
-      bool pass = true;
-      if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
-      const float vn = (metric == 0) ? csr_vnorms[r] : 0.f;
-
-      float mykey = 0.f;  // lane j ends up holding query j's key for row r
-      for (int32_t j = 0; j < qt; j++) {
-        const float4* q4 = (const float4*)(smem + (size_t)j * d);
-        float acc = 0.f;
-#pragma unroll
-        for (int i = 0; i < PER_LANE; i++) {
-          int idx = lane + i * WAVE;
-          if (idx < d4) {
-            float4 a = cur[i], b = q4[idx];
-            acc += a.x * b.x + a.y * b.y + a.z * b.z + a.w * b.w;
-          }
-        }
-        for (int off = 32; off; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
-        // L2 key = vnorm - 2*dot (+qnorm at emit); IP key = -dot
-        float key = (metric == 0) ? vn - 2.0f * acc : -acc;
-        if (lane == j) mykey = key;
-      }
-      if (lane < qt) {  // one vector-store wave-instruction, qt active lanes
-        uint64_t c = pass ? pack_cand(mykey, (uint32_t)r) : kCandEmpty;
-        cand[my_cbase + r] = c;
-      }
-#pragma unroll
-      for (int i = 0; i < PER_LANE; i++) cur[i] = nxt[i];
-    }
-  }
-}
-
 // ---------- range search (radius) ----------
 // candidates (packed u64) below a per-query threshold key.  Strict '<'
 // matches faiss RangeSearch (L2: dist < radius; IP: score > radius).
@@ -1204,41 +1095,6 @@ void fill_units(hipStream_t s, const int32_t* unit_offsets,
                 uint32_t* units, int32_t /*total*/) {
   hipLaunchKernelGGL(k_fill_units, dim3(ceil_div(nlist, 128)), dim3(128), 0,
                      s, unit_offsets, unit_counts, nlist, units);
-}
-
-void ivf_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
-              const int64_t* csr_offsets, const float* csr_vectors,
-              const float* csr_vnorms, const int64_t* /*csr_ids*/,
-              const float* queries, const float* /*qnorms*/, int32_t d,
-              const int32_t* inv_offsets, const int32_t* inv_q,
-              const int32_t* inv_rank, const int64_t* qp_off,
-              const int64_t* q_cand_base, int metric,
-              const uint32_t* bitmap, int32_t chunk_rows, int32_t nprobe,
-              int32_t qt_max, uint64_t* cand) {
-  if (!n_units) return;
-  size_t lds = (size_t)qt_max * d * 4 + (size_t)qt_max * 8;
-  const int per_lane = (d / 4 + WAVE - 1) / WAVE;  // ceil(d/256)
-#define DG_SCAN_CASE(PL)                                                     \
-  case PL:                                                                   \
-    hipLaunchKernelGGL(k_ivf_scan<PL>, dim3((uint32_t)n_units), dim3(256),   \
-                       lds, s, units, csr_offsets, csr_vectors, csr_vnorms,  \
-                       queries, d, inv_offsets, inv_q, inv_rank, qp_off,     \
-                       q_cand_base, nprobe, metric, bitmap, chunk_rows,      \
-                       qt_max, cand);                                        \
-    break;
-  switch (per_lane) {
-    DG_SCAN_CASE(1)
-    DG_SCAN_CASE(2)
-    DG_SCAN_CASE(3)
-    DG_SCAN_CASE(4)
-    DG_SCAN_CASE(5)
-    DG_SCAN_CASE(6)
-    DG_SCAN_CASE(7)
-    DG_SCAN_CASE(8)
-    default:
-      break;  // host validates d <= 2048 for IVF scan
-  }
-#undef DG_SCAN_CASE
 }
 
 void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
