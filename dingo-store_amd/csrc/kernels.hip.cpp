@@ -521,7 +521,7 @@ __global__ void k_tombstone(const int64_t* __restrict__ pos, int64_t n,
 // Column loads are 16 B/lane coalesced; query element broadcasts come from
 // LDS.  Unused query slots are zero-staged so the inner loop is branch-free.
 template <int QTM, int RPL>  // queries per tile, rows per lane
-__global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
+__device__ __forceinline__ void ivf_scan_col_body(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
     const float* __restrict__ tvec, const float* __restrict__ vnorms,
@@ -812,6 +812,44 @@ __global__ void __launch_bounds__(256, 4) k_ivfpq_scan(
       }
     }
   }
+}
+
+// wrappers: same body, different register budgets.  The (256,1) form lets
+// the allocator use 209 VGPRs (2 blocks/CU, LDS would allow 3); the
+// (256,3) form caps at 168 VGPRs for 3 waves/SIMD at the cost of possible
+// cold spills -- A/B'd via DG_SCAN_VARIANT.
+template <int QTM, int RPL>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  ivf_scan_col_body<QTM, RPL>(units, csr_offsets, chunk_off, chunk_base,
+                              tvec, vnorms, queries, d, inv_offsets, inv_q,
+                              inv_rank, qp_off, q_cand_base, nprobe, metric,
+                              bitmap, chunk_rows, cand);
+}
+
+template <int QTM, int RPL>
+__global__ void __launch_bounds__(256, 3) k_ivf_scan_col_hi(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  ivf_scan_col_body<QTM, RPL>(units, csr_offsets, chunk_off, chunk_base,
+                              tvec, vnorms, queries, d, inv_offsets, inv_q,
+                              inv_rank, qp_off, q_cand_base, nprobe, metric,
+                              bitmap, chunk_rows, cand);
 }
 
 // tiled row-major -> column-major chunk transpose (finalize step).
@@ -1127,6 +1165,17 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     case 3: DG_SCAN_LAUNCH(8, 8); break;
     case 4: DG_SCAN_LAUNCH(4, 4); break;
     case 5: DG_SCAN_LAUNCH(12, 4); break;
+    case 6: {
+      constexpr int QTM = 16;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_col_hi<QTM, 4>),
+                         dim3((uint32_t)n_units), dim3(256), lds, s, units,
+                         csr_offsets, chunk_off, chunk_base, tvec, vnorms,
+                         queries, d, inv_offsets, inv_q, inv_rank, qp_off,
+                         q_cand_base, nprobe, metric, bitmap, chunk_rows,
+                         cand);
+      break;
+    }
     default: DG_SCAN_LAUNCH(16, 4); break;
   }
 #undef DG_SCAN_LAUNCH
