@@ -70,9 +70,27 @@ def build_index(args, rank, world, device):
     n, d, nlist = args.n, args.d, args.nlist
     r0 = rank * n // world
     r1 = (rank + 1) * n // world
-    kind = dg.IVF_PQ if args.kind == "ivf_pq" else dg.IVF_FLAT
+    kind = {"ivf_pq": dg.IVF_PQ, "flat": dg.FLAT,
+            "ivf_flat": dg.IVF_FLAT}[args.kind]
     idx = dg.Index(kind, dg.L2, d, nlist=nlist, m=args.m,
                    device=device.index, reserve=(r1 - r0))
+    if kind == dg.FLAT:  # cfg B: exact scan, no train/coarse stage
+        t0 = time.time()
+        pos = r0
+        while pos < r1:
+            c = pos // GEN_CHUNK
+            c_start = c * GEN_CHUNK
+            rows_all = min(GEN_CHUNK, n - c_start)
+            lo, hi = pos - c_start, min(rows_all, r1 - c_start)
+            chunk = gen_chunk_device(args.seed, c, rows_all, d, device)
+            part = chunk[lo:hi].contiguous()
+            ids = np.arange(c_start + lo, c_start + hi, dtype=np.int64)
+            idx.add_device(ids, part.data_ptr(), part.shape[0])
+            del chunk, part
+            pos = c_start + hi
+        torch.cuda.synchronize()
+        log(rank, f"add {r1-r0} rows {time.time()-t0:.1f}s")
+        return idx, (r0, r1)
 
     # ---- train on rank 0 (first 256*nlist rows of the global set), then
     # broadcast centroids (mirrors TrainForBuild + snapshot install roles)
@@ -218,6 +236,33 @@ def cpu_baseline(idx, args, q_host):
             "cores": cores, "kind": "port", "sample": sample}
 
 
+def cpu_baseline_flat(idx, args, q_host):
+    """Flat (cfg B) CPU baseline: oracle exhaustive scan, host cores."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import pyoracle as orc
+    n, d, k = args.n, args.d, args.k
+    base = np.empty((n, d), np.float32)
+    dev = torch.device("cuda:0")
+    for c in range((n + GEN_CHUNK - 1) // GEN_CHUNK):
+        rows = min(GEN_CHUNK, n - c * GEN_CHUNK)
+        t = gen_chunk_device(args.seed, c, rows, d, dev)
+        base[c * GEN_CHUNK:c * GEN_CHUNK + rows] = t.cpu().numpy()
+        del t
+    t0 = time.time()
+    nq_probe = 4
+    orc.flat_search(0, base, q_host[:nq_probe], k, fast=True)
+    per_q = (time.time() - t0) / nq_probe
+    nq_sample = int(min(len(q_host), max(4, 15.0 / max(per_q, 1e-6))))
+    t0 = time.time()
+    orc.flat_search(0, base, q_host[:nq_sample], k, fast=True)
+    dt = time.time() - t0
+    cores = os.cpu_count()
+    return {"value": round(nq_sample / dt, 2), "unit": "queries/s",
+            "cores": cores, "kind": "port",
+            "sample": f"{nq_sample} queries exhaustive over {n} rows, "
+                      f"{dt:.1f}s, OpenMP {cores} cores"}
+
+
 def read_traffic(workload_name):
     """Per-launch HBM traffic measured by a separate rocprofv3 --pmc run
     (profiles/pmc_traffic.json, written by profiling scripts); null if no
@@ -245,7 +290,7 @@ def main():
     ap.add_argument("--nprobe", type=int, default=32)
     ap.add_argument("--k", type=int, default=10)
     ap.add_argument("--seed", type=int, default=4244)
-    ap.add_argument("--kind", choices=["ivf_flat", "ivf_pq"],
+    ap.add_argument("--kind", choices=["ivf_flat", "ivf_pq", "flat"],
                     default="ivf_flat")
     ap.add_argument("--m", type=int, default=96)  # cfg D subquantizers
     ap.add_argument("--quick", action="store_true",
@@ -310,7 +355,7 @@ def main():
         total_ms.append(st["last_total_ms"])
         alg_bytes.append(st["last_scan_bytes_algorithmic"])
 
-    recall = None
+    recall = 1.0 if args.kind == "flat" else None  # Flat IS the exact scan
     if not args.no_recall and args.kind == "ivf_flat":
         # (PQ exact-sweep GT at cfg D would need a 3.2 TB candidate buffer;
         # PQ recall is covered by the parity tests at tractable sizes)
@@ -318,10 +363,12 @@ def main():
 
     cpu = None
     if (rank == 0 and world == 1 and not args.no_cpu_baseline
-            and args.kind == "ivf_flat"):
+            and args.kind in ("ivf_flat", "flat")):
         q_host = q.cpu().numpy()
         try:
-            cpu = cpu_baseline(idx, args, q_host)
+            cpu = (cpu_baseline_flat(idx, args, q_host)
+                   if args.kind == "flat" else
+                   cpu_baseline(idx, args, q_host))
         except Exception as ex:
             log(rank, f"cpu baseline failed: {ex}")
 
@@ -332,15 +379,18 @@ def main():
         peak_gbps = 8000.0  # HBM3E spec (measured ceiling ~6300 GB/s,
         # MI355X_MICROARCH.md); fraction vs spec per §8d
         achieved = alg / (scan_s * 1e6) if scan_s > 0 else 0.0
-        kind_name = ("IVF-PQ m=" + str(args.m) if args.kind == "ivf_pq"
-                     else "IVF-Flat")
+        kind_name = {"ivf_pq": "IVF-PQ m=" + str(args.m), "flat": "Flat",
+                     "ivf_flat": "IVF-Flat"}[args.kind]
         workload_name = (f"{kind_name} {args.n//10**6}M x {args.d} fp32 "
-                         f"nlist={args.nlist} nprobe={nprobe} "
-                         f"batch={nq} k={k}")
+                         + ("" if args.kind == "flat"
+                            else f"nlist={args.nlist} nprobe={nprobe} ")
+                         + f"batch={nq} k={k}")
         out = {
-            "metric": ("QPS @ recall@10, IVF-Flat 10Mx768 nprobe=32"
-                       if args.kind == "ivf_flat" else
-                       "QPS, IVF-PQ (BASELINE cfg D)"),
+            "metric": {"ivf_flat": "QPS @ recall@10, IVF-Flat 10Mx768 "
+                                   "nprobe=32",
+                       "ivf_pq": "QPS, IVF-PQ (BASELINE cfg D)",
+                       "flat": "QPS, Flat exact (BASELINE cfg B)"}[
+                           args.kind],
             "value": round(qps, 1),
             "unit": "queries/s",
             "n_gpus": world,
@@ -362,7 +412,9 @@ def main():
                                "top-k" if world > 1 else "single GPU",
             },
             "roofline": {
-                "bound": "hbm",
+                # cfg B is compute-bound (128 FLOP/B > fp32 ridge,
+                # SURVEY.md §8d); the IVF/PQ scans are HBM-bound
+                "bound": "mfma" if args.kind == "flat" else "hbm",
                 "achieved": round(achieved, 1),
                 "peak": peak_gbps,
                 "unit": "GB/s",
