@@ -376,9 +376,19 @@ def main():
         qps = nq * args.steps / elapsed
         scan_s = float(np.mean(scan_ms))
         alg = int(np.mean(alg_bytes))
-        peak_gbps = 8000.0  # HBM3E spec (measured ceiling ~6300 GB/s,
-        # MI355X_MICROARCH.md); fraction vs spec per §8d
-        achieved = alg / (scan_s * 1e6) if scan_s > 0 else 0.0
+        if args.kind == "flat":
+            # compute-bound: exact-scan FLOPs vs the fp32 MFMA peak
+            # (157.3 TF, MI355X_MICROARCH.md; scan time includes the
+            # per-chunk select pass)
+            flops = 2.0 * nq * args.n * args.d
+            peak = 157.3e12
+            achieved = flops / (scan_s * 1e-3) if scan_s > 0 else 0.0
+            unit = "FLOP/s"
+        else:
+            peak = 8000.0  # HBM3E spec GB/s (measured ceiling ~6300,
+            # MI355X_MICROARCH.md); fraction vs spec per §8d
+            achieved = alg / (scan_s * 1e6) if scan_s > 0 else 0.0
+            unit = "GB/s"
         kind_name = {"ivf_pq": "IVF-PQ m=" + str(args.m), "flat": "Flat",
                      "ivf_flat": "IVF-Flat"}[args.kind]
         workload_name = (f"{kind_name} {args.n//10**6}M x {args.d} fp32 "
@@ -416,9 +426,9 @@ def main():
                 # SURVEY.md §8d); the IVF/PQ scans are HBM-bound
                 "bound": "mfma" if args.kind == "flat" else "hbm",
                 "achieved": round(achieved, 1),
-                "peak": peak_gbps,
-                "unit": "GB/s",
-                "frac": round(achieved / peak_gbps, 4),
+                "peak": peak,
+                "unit": unit,
+                "frac": round(achieved / peak, 4),
                 "traffic": read_traffic(workload_name),
                 "detail": {
                     "scan_ms_per_launch": round(scan_s, 3),
