@@ -731,7 +731,7 @@ __global__ void k_build_S(const float* __restrict__ centroids,
 // vector and walks m = 0..M-1 sequentially, so all lanes of a wave gather
 // within the same 1 KB rows of T[q] and S[l] (L1-resident after first
 // touch).  No cross-lane reduction.
-__global__ void __launch_bounds__(256, 4) k_ivfpq_scan(
+__global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const uint8_t* __restrict__ csr_codes, const float* __restrict__ S,
     const float* __restrict__ T, const float* __restrict__ coarse_dots,
@@ -777,31 +777,32 @@ __global__ void __launch_bounds__(256, 4) k_ivfpq_scan(
       for (int32_t v = lane; v < tn; v += WAVE) {
         const uint32_t* code4 = (const uint32_t*)(lds_codes + (size_t)v * M);
         float acc = 0.f;
-        // m unrolled by 4 (one LDS dword = 4 codes; 8 independent gather
-        // chains in flight instead of one serial byte->gather->add chain)
+        // m unrolled by 4 (one LDS dword = 4 codes).  Table bases advance
+        // by wave-uniform pointer bumps (SALU) so each gather is one
+        // 32-bit shift + saddr load, with the 256/512/768 row offsets
+        // folded into instruction immediates — 64-bit per-gather address
+        // math was half the inner loop's instructions.
         if (metric == 0) {
+          const float* Sm = Sl;
+          const float* Tm = Tq;
           for (int32_t m4 = 0; m4 < M / 4; m4++) {
             const uint32_t cw = code4[m4];
-            const int32_t m = m4 * 4;
-            float s0 = Sl[(m + 0) * 256 + (cw & 255)];
-            float s1 = Sl[(m + 1) * 256 + ((cw >> 8) & 255)];
-            float s2 = Sl[(m + 2) * 256 + ((cw >> 16) & 255)];
-            float s3 = Sl[(m + 3) * 256 + (cw >> 24)];
-            float t0 = Tq[(m + 0) * 256 + (cw & 255)];
-            float t1 = Tq[(m + 1) * 256 + ((cw >> 8) & 255)];
-            float t2 = Tq[(m + 2) * 256 + ((cw >> 16) & 255)];
-            float t3 = Tq[(m + 3) * 256 + (cw >> 24)];
-            acc += (s0 + s1 + s2 + s3) - 2.0f * (t0 + t1 + t2 + t3);
+            const uint32_t c0 = cw & 255, c1 = (cw >> 8) & 255,
+                           c2 = (cw >> 16) & 255, c3 = cw >> 24;
+            acc += (Sm[c0] + Sm[256 + c1] + Sm[512 + c2] + Sm[768 + c3]) -
+                   2.0f * (Tm[c0] + Tm[256 + c1] + Tm[512 + c2] +
+                           Tm[768 + c3]);
+            Sm += 1024;
+            Tm += 1024;
           }
           acc -= 2.0f * dot;  // + qnorm at emit
         } else {
+          const float* Tm = Tq;
           for (int32_t m4 = 0; m4 < M / 4; m4++) {
             const uint32_t cw = code4[m4];
-            const int32_t m = m4 * 4;
-            acc += Tq[(m + 0) * 256 + (cw & 255)] +
-                   Tq[(m + 1) * 256 + ((cw >> 8) & 255)] +
-                   Tq[(m + 2) * 256 + ((cw >> 16) & 255)] +
-                   Tq[(m + 3) * 256 + (cw >> 24)];
+            acc += Tm[cw & 255] + Tm[256 + ((cw >> 8) & 255)] +
+                   Tm[512 + ((cw >> 16) & 255)] + Tm[768 + (cw >> 24)];
+            Tm += 1024;
           }
           acc = -(acc + dot);  // IP key = -score
         }
