@@ -766,7 +766,11 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
     }
     __syncthreads();
 
-    // waves split the probing queries; lanes split the tile's vectors
+    // waves split the probing queries; each LANE owns 4 of the 256 staged
+    // vectors (v, v+64, v+128, v+192), so every (wave, query, m) visit of a
+    // 1 KB table row is consumed by all 256 vectors — without this, 3 KB of
+    // S/T row traffic per (vector, query) pair is the bottleneck (S at
+    // cfg D is 1.6 GB, beyond L3).
     for (int32_t qi = wave_id; qi < nql; qi += blockDim.x / WAVE) {
       const int32_t q = inv_q[iq0 + qi];
       const int32_t rank = inv_rank[iq0 + qi];
@@ -774,42 +778,58 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
       const float dot = coarse_dots[(size_t)q * nlist + list];
       const int64_t cb0 = q_cand_base[q] +
                           qp_off[(int64_t)q * nprobe + rank] - list_start;
-      for (int32_t v = lane; v < tn; v += WAVE) {
-        const uint32_t* code4 = (const uint32_t*)(lds_codes + (size_t)v * M);
-        float acc = 0.f;
-        // m unrolled by 4 (one LDS dword = 4 codes).  Table bases advance
-        // by wave-uniform pointer bumps (SALU) so each gather is one
-        // 32-bit shift + saddr load, with the 256/512/768 row offsets
-        // folded into instruction immediates — 64-bit per-gather address
-        // math was half the inner loop's instructions.
-        if (metric == 0) {
-          const float* Sm = Sl;
-          const float* Tm = Tq;
-          for (int32_t m4 = 0; m4 < M / 4; m4++) {
-            const uint32_t cw = code4[m4];
+      constexpr int RPV = 4;  // TILE = RPV * WAVE
+      float acc[RPV];
+      const uint32_t* code4[RPV];
+#pragma unroll
+      for (int r = 0; r < RPV; r++) {
+        acc[r] = 0.f;
+        code4[r] =
+            (const uint32_t*)(lds_codes + (size_t)(lane + r * WAVE) * M);
+      }
+      if (metric == 0) {
+        const float* Sm = Sl;
+        const float* Tm = Tq;
+        for (int32_t m4 = 0; m4 < M / 4; m4++) {
+#pragma unroll
+          for (int r = 0; r < RPV; r++) {
+            const uint32_t cw = code4[r][m4];
             const uint32_t c0 = cw & 255, c1 = (cw >> 8) & 255,
                            c2 = (cw >> 16) & 255, c3 = cw >> 24;
-            acc += (Sm[c0] + Sm[256 + c1] + Sm[512 + c2] + Sm[768 + c3]) -
-                   2.0f * (Tm[c0] + Tm[256 + c1] + Tm[512 + c2] +
-                           Tm[768 + c3]);
-            Sm += 1024;
-            Tm += 1024;
+            acc[r] +=
+                (Sm[c0] + Sm[256 + c1] + Sm[512 + c2] + Sm[768 + c3]) -
+                2.0f * (Tm[c0] + Tm[256 + c1] + Tm[512 + c2] +
+                        Tm[768 + c3]);
           }
-          acc -= 2.0f * dot;  // + qnorm at emit
-        } else {
-          const float* Tm = Tq;
-          for (int32_t m4 = 0; m4 < M / 4; m4++) {
-            const uint32_t cw = code4[m4];
-            acc += Tm[cw & 255] + Tm[256 + ((cw >> 8) & 255)] +
-                   Tm[512 + ((cw >> 16) & 255)] + Tm[768 + (cw >> 24)];
-            Tm += 1024;
-          }
-          acc = -(acc + dot);  // IP key = -score
+          Sm += 1024;
+          Tm += 1024;
         }
-        const int64_t r = t0 + v;
-        bool pass = true;
-        if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
-        cand[cb0 + r] = pass ? pack_cand(acc, (uint32_t)r) : kCandEmpty;
+#pragma unroll
+        for (int r = 0; r < RPV; r++) acc[r] -= 2.0f * dot;  // +qnorm@emit
+      } else {
+        const float* Tm = Tq;
+        for (int32_t m4 = 0; m4 < M / 4; m4++) {
+#pragma unroll
+          for (int r = 0; r < RPV; r++) {
+            const uint32_t cw = code4[r][m4];
+            acc[r] += Tm[cw & 255] + Tm[256 + ((cw >> 8) & 255)] +
+                      Tm[512 + ((cw >> 16) & 255)] + Tm[768 + (cw >> 24)];
+          }
+          Tm += 1024;
+        }
+#pragma unroll
+        for (int r = 0; r < RPV; r++) acc[r] = -(acc[r] + dot);
+      }
+#pragma unroll
+      for (int r = 0; r < RPV; r++) {
+        const int32_t v = lane + r * WAVE;
+        if (v < tn) {
+          const int64_t rw = t0 + v;
+          bool pass = true;
+          if (bitmap) pass = (bitmap[rw >> 5] >> (rw & 31)) & 1;
+          cand[cb0 + rw] =
+              pass ? pack_cand(acc[r], (uint32_t)rw) : kCandEmpty;
+        }
       }
     }
   }
