@@ -617,15 +617,21 @@ __device__ __forceinline__ void ivf_scan_col_body(
           }
         }
       };
+      static_assert(U % 4 == 0, "U");
       auto compute_block = [&](float (&c)[U][RPL], int32_t ib) {
 #pragma unroll
         for (int j = 0; j < QTM; j++) {
-          const float4 qv4 = *(const float4*)(smem + (size_t)j * d + ib);
-          const float qv[4] = {qv4.x, qv4.y, qv4.z, qv4.w};
 #pragma unroll
-          for (int uu = 0; uu < U; uu++)
+          for (int u4 = 0; u4 < U / 4; u4++) {
+            const float4 qv4 =
+                *(const float4*)(smem + (size_t)j * d + ib + u4 * 4);
+            const float qv[4] = {qv4.x, qv4.y, qv4.z, qv4.w};
 #pragma unroll
-            for (int x = 0; x < RPL; x++) acc[j][x] += c[uu][x] * qv[uu];
+            for (int uu = 0; uu < 4; uu++)
+#pragma unroll
+              for (int x = 0; x < RPL; x++)
+                acc[j][x] += c[u4 * 4 + uu][x] * qv[uu];
+          }
         }
       };
       // 2-deep rotation (3-deep costs a wave of occupancy and regresses)
