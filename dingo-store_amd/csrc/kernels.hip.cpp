@@ -591,7 +591,7 @@ __device__ __forceinline__ void ivf_scan_col_body(
       // block i+U's loads are in flight (double-buffered column registers —
       // waiting within the issuing iteration exposes full HBM latency).
       // Query elements read 4-at-a-time from aligned LDS.
-      constexpr int U = 8;
+      constexpr int U = 4;
       static_assert(RPL == 1 || RPL == 2 || RPL % 4 == 0, "RPL");
       float ca[U][RPL], cb[U][RPL];
       const float* colp = col + rr0;
