@@ -926,8 +926,11 @@ static dg_status finalize_csr(dg_index* ix) {
                         (size_t)ix->total_chunks * 8;
     if ((st = dbuf_reserve(ix->d_chunk_meta, meta_bytes, ix->stream,
                            false)) != DG_OK ||
-        (st = dbuf_reserve(ix->d_csr_t, (size_t)t_elems * 4, ix->stream,
-                           false)) != DG_OK) {
+        // +1 chunk of slack: the glds scan's tail quarters issue padded
+        // reads past the last chunk (results discarded at emit)
+        (st = dbuf_reserve(ix->d_csr_t,
+                           ((size_t)t_elems + (size_t)d * CR) * 4,
+                           ix->stream, false)) != DG_OK) {
       dbuf_free(rm_tmp);
       return st;
     }

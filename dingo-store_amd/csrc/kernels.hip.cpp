@@ -841,6 +841,156 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
   }
 }
 
+// ---------- glds scan (v3 experiment, DG_SCAN_VARIANT=7) ----------
+// Column data streams HBM -> LDS via global_load_lds into a per-wave
+// private ring (each wave owns a 256-row quarter of the chunk, so no
+// cross-wave hand-off and no barriers in the steady loop; counted vmcnt
+// waits only).  Consumers read columns from LDS (conflict-free b128) and
+// accumulate lane-private dots as in v2.  1 block/CU (LDS-bound).
+template <int QTM, int DEPTH>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_glds(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  // LDS: [QTM*d] query tile | per-wave rings [4][DEPTH][4*256] | cbase[QTM]
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  constexpr int SLOT = 4 * 256;  // 4 dims x 256 rows (f32)
+  float* ring_all = smem + (size_t)QTM * d;
+  int64_t* cbase = (int64_t*)(ring_all + 4 * DEPTH * SLOT);
+
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* col = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+  const int64_t row0 = list_start + (int64_t)chunk * chunk_rows;
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  __builtin_assume(d % 4 == 0 && d > 0);
+  float* ring = ring_all + wave_id * DEPTH * SLOT;
+  const int32_t wrow0 = wave_id * 256;          // this wave's row quarter
+  if (wrow0 >= nrows_pad) {
+    // idle quarter still participates in the staging barriers below
+  }
+  const int32_t rr0 = wrow0 + lane * 4;         // this lane's 4 rows
+  const int nslots = d / 4;                     // slots per full pass
+
+  for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
+    const int32_t qt = min(QTM, nql - t0);
+    __syncthreads();
+    for (int32_t j = 0; j < qt; j++) {
+      int32_t q = inv_q[iq0 + t0 + j];
+      const float4* src = (const float4*)(queries + (size_t)q * d);
+      float4* dst = (float4*)(smem + (size_t)j * d);
+      for (int i = threadIdx.x; i < d / 4; i += blockDim.x) dst[i] = src[i];
+    }
+    for (size_t i = (size_t)qt * d + threadIdx.x; i < (size_t)QTM * d;
+         i += blockDim.x)
+      smem[i] = 0.f;
+    if (threadIdx.x < QTM) {
+      int32_t j = threadIdx.x;
+      if (j < qt) {
+        int32_t q = inv_q[iq0 + t0 + j];
+        int32_t rank = inv_rank[iq0 + t0 + j];
+        cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                   list_start;
+      } else {
+        cbase[j] = 0;
+      }
+    }
+    __syncthreads();
+    if (wrow0 < nrows_pad) {
+      float acc[QTM][4];
+#pragma unroll
+      for (int j = 0; j < QTM; j++)
+#pragma unroll
+        for (int x = 0; x < 4; x++) acc[j][x] = 0.f;
+
+      // issue one slot: 4 glds of 1 KiB (dim column slice for this quarter)
+      auto issue = [&](int32_t slot_i /* dim block index */) {
+        float* ldst = ring + (slot_i % DEPTH) * SLOT;
+#pragma unroll
+        for (int u = 0; u < 4; u++) {
+          const float* gsrc =
+              col + (size_t)(slot_i * 4 + u) * nrows_pad + wrow0 + lane * 4;
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) uint32_t*)gsrc,
+              (__attribute__((address_space(3))) uint32_t*)(ldst + u * 256),
+              16, 0, 0);
+        }
+      };
+      auto consume = [&](int32_t slot_i) {
+        const float* sl = ring + (slot_i % DEPTH) * SLOT + lane * 4;
+#pragma unroll
+        for (int u = 0; u < 4; u++) {
+          const float4 c4 = *(const float4*)(sl + u * 256);
+#pragma unroll
+          for (int j = 0; j < QTM; j++) {
+            const float qv = smem[(size_t)j * d + slot_i * 4 + u];
+            acc[j][0] += c4.x * qv;
+            acc[j][1] += c4.y * qv;
+            acc[j][2] += c4.z * qv;
+            acc[j][3] += c4.w * qv;
+          }
+        }
+      };
+      const int pre = min(DEPTH, nslots);
+      for (int s = 0; s < pre; s++) issue(s);
+      for (int s = 0; s < nslots; s++) {
+        // wait until slot s's 4 glds landed; leave the younger slots'
+        // glds in flight: outstanding after wait = 4*(issued - s - 1).
+        // asm form with memory clobber so ds_reads can't hoist above it
+        // (counted waits only — a vmcnt(0) here would drain the ring).
+        const int n_out = 4 * (min(nslots, s + DEPTH) - s - 1);
+        switch (n_out) {
+          case 0: asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); break;
+          case 4: asm volatile("s_waitcnt vmcnt(4)" ::: "memory"); break;
+          case 8: asm volatile("s_waitcnt vmcnt(8)" ::: "memory"); break;
+          case 12: asm volatile("s_waitcnt vmcnt(12)" ::: "memory"); break;
+          case 16: asm volatile("s_waitcnt vmcnt(16)" ::: "memory"); break;
+          case 20: asm volatile("s_waitcnt vmcnt(20)" ::: "memory"); break;
+          default: asm volatile("s_waitcnt vmcnt(24)" ::: "memory"); break;
+        }
+        consume(s);
+        if (s + DEPTH < nslots) issue(s + DEPTH);
+      }
+      (void)pre;
+      // emit (guards, not breaks — see v2)
+#pragma unroll
+      for (int j = 0; j < QTM; j++) {
+        if (j < qt) {
+          const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+          for (int x = 0; x < 4; x++) {
+            const int32_t rl = rr0 + x;
+            if (rl < nrows) {
+              const int64_t r = row0 + rl;
+              bool pass = true;
+              if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+              float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                        : -acc[j][x];
+              cand[cb + x] = pass ? pack_cand(key, (uint32_t)r)
+                                  : kCandEmpty;
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
 // wrappers: same body, different register budgets.  The (256,1) form lets
 // the allocator use 209 VGPRs (2 blocks/CU, LDS would allow 3); the
 // (256,3) form caps at 168 VGPRs for 3 waves/SIMD at the cost of possible
@@ -1192,6 +1342,17 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     case 3: DG_SCAN_LAUNCH(8, 8); break;
     case 4: DG_SCAN_LAUNCH(4, 4); break;
     case 5: DG_SCAN_LAUNCH(12, 4); break;
+    case 7: {
+      constexpr int QTM = 16, DEPTH = 6;
+      size_t lds = (size_t)QTM * d * 4 + 4 * DEPTH * 4 * 256 * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_glds<QTM, DEPTH>),
+                         dim3((uint32_t)n_units), dim3(256), lds, s, units,
+                         csr_offsets, chunk_off, chunk_base, tvec, vnorms,
+                         queries, d, inv_offsets, inv_q, inv_rank, qp_off,
+                         q_cand_base, nprobe, metric, bitmap, chunk_rows,
+                         cand);
+      break;
+    }
     case 6: {
       constexpr int QTM = 16;
       size_t lds = (size_t)QTM * d * 4 + QTM * 8;
