@@ -623,9 +623,12 @@ __device__ __forceinline__ void ivf_scan_col_body(
         for (int j = 0; j < QTM; j++) {
 #pragma unroll
           for (int u4 = 0; u4 < U / 4; u4++) {
-            const float4 qv4 =
-                *(const float4*)(smem + (size_t)j * d + ib + u4 * 4);
-            const float qv[4] = {qv4.x, qv4.y, qv4.z, qv4.w};
+            // two b64 LDS reads (2cy/8B) instead of a float4 that lowers
+            // to ds_read2_b32 (4cy/8B)
+            const float* qp = smem + (size_t)j * d + ib + u4 * 4;
+            const float2 qa = *(const float2*)qp;
+            const float2 qb = *(const float2*)(qp + 2);
+            const float qv[4] = {qa.x, qa.y, qb.x, qb.y};
 #pragma unroll
             for (int uu = 0; uu < 4; uu++)
 #pragma unroll
