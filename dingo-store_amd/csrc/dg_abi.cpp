@@ -231,7 +231,7 @@ extern "C" void dg_index_destroy(dg_index* ix) {
         &ix->d_cnorms, &ix->d_csr_offsets, &ix->d_csr_vectors, &ix->d_csr_ids,
         &ix->d_csr_vnorms, &ix->d_csr_t, &ix->d_chunk_meta, &ix->d_list_mask,
         &ix->d_codebooks, &ix->d_codes, &ix->d_csr_codes, &ix->d_S,
-        &ix->d_cb_norms, &ix->ws_T, &ix->ws_queries, &ix->ws_qnorms,
+        &ix->d_cb_norms, &ix->ws_T, &ix->ws_Tf32, &ix->ws_queries, &ix->ws_qnorms,
         &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
         &ix->ws_small, &ix->ws_topk})
     dbuf_free(*b);
@@ -449,14 +449,14 @@ static dg_status finish_pq_tables(dg_index* ix) {
   const int32_t dsub = d / M;
   const int32_t nlist = ix->desc.nlist;
   dg_status st;
-  if ((st = dbuf_reserve(ix->d_S, (size_t)nlist * M * 256 * 4, ix->stream,
+  if ((st = dbuf_reserve(ix->d_S, (size_t)nlist * M * 256 * 2, ix->stream,
                          false)) != DG_OK ||
       (st = dbuf_reserve(ix->d_cb_norms, (size_t)M * 256 * 4, ix->stream,
                          false)) != DG_OK)
     return st;
   dgk::build_S(ix->stream, (const float*)ix->d_centroids.p,
                (const float*)ix->d_codebooks.p, nlist, M, dsub, d,
-               (float*)ix->d_S.p);
+               (__half*)ix->d_S.p);
   dgk::row_norms(ix->stream, (const float*)ix->d_codebooks.p,
                  (int64_t)M * 256, dsub, (float*)ix->d_cb_norms.p);
   return DG_OK;
@@ -1406,8 +1406,10 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
       // subspace) then gather-scan codes
       const int32_t M = ix->desc.pq_m;
       const int32_t dsub = d / M;
-      if ((st = dbuf_reserve(ix->ws_T, (size_t)nq * M * 256 * 4, ix->stream,
-                             false)) != DG_OK) {
+      if ((st = dbuf_reserve(ix->ws_T, (size_t)nq * M * 256 * 2, ix->stream,
+                             false)) != DG_OK ||
+          (st = dbuf_reserve(ix->ws_Tf32, (size_t)nq * M * 256 * 4,
+                             ix->stream, false)) != DG_OK) {
         dbuf_free(d_fids);
         dbuf_free(ws_bitmap);
         return st;
@@ -1417,7 +1419,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
               ix->blas, rocblas_operation_transpose, rocblas_operation_none,
               256, (rocblas_int)nq, dsub, &one,
               (const float*)ix->d_codebooks.p, dsub, (int64_t)256 * dsub, dq,
-              d, (int64_t)dsub, &zero, (float*)ix->ws_T.p,
+              d, (int64_t)dsub, &zero, (float*)ix->ws_Tf32.p,
               (rocblas_int)(M * 256), (int64_t)256,
               M) != rocblas_status_success) {
         dg_set_error("T build sgemm failed");
@@ -1425,10 +1427,12 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
         dbuf_free(ws_bitmap);
         return DG_EINTERNAL;
       }
+      dgk::f32_to_f16(ix->stream, (const float*)ix->ws_Tf32.p,
+                      (int64_t)nq * M * 256, (__half*)ix->ws_T.p);
       dgk::ivfpq_scan(ix->stream, units, total_units,
                       (const int64_t*)ix->d_csr_offsets.p,
                       (const uint8_t*)ix->d_csr_codes.p,
-                      (const float*)ix->d_S.p, (const float*)ix->ws_T.p,
+                      (const __half*)ix->d_S.p, (const __half*)ix->ws_T.p,
                       (const float*)ix->ws_dots.p, nlist, M, inv_offsets32,
                       inv_q, inv_rank, qp_off, q_cand_base, np, metric,
                       d_bitmap, chunk_rows, (uint64_t*)ix->ws_cand.p);

@@ -4,6 +4,7 @@
 // fallback — entry points fail with DG_ENOGPU when no device is present.
 #pragma once
 
+#include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 #include <rocblas/rocblas.h>
 
@@ -106,9 +107,10 @@ struct dg_index {
   dg_dbuf d_codes;          // arrival [ntotal x M] u8 (raw vectors are NOT
                             // kept for PQ; d_vectors stays empty)
   dg_dbuf d_csr_codes;      // grouped [ntotal x M] u8
-  dg_dbuf d_S;              // [nlist][M][256] f32 ||c_m + cb||^2
+  dg_dbuf d_S;              // [nlist][M][256] f16 ||c_m + cb||^2
   dg_dbuf d_cb_norms;       // [M][256] f32 codebook entry norms (encode)
-  dg_dbuf ws_T;             // per-batch [nq][M][256] f32 q_sub . cb
+  dg_dbuf ws_T;             // per-batch [nq][M][256] f16 q_sub . cb
+  dg_dbuf ws_Tf32;          // f32 GEMM output before f16 convert
 
   // optional list ownership mask (multi-GPU list sharding)
   dg_dbuf d_list_mask;      // [nlist] u8, empty = all owned
@@ -221,10 +223,11 @@ void set_code(hipStream_t s, const int32_t* amin, int64_t n, int32_t m,
 void gather_codes(hipStream_t s, const uint8_t* src, const uint32_t* perm,
                   int64_t n, int32_t M, uint8_t* dst);
 void build_S(hipStream_t s, const float* centroids, const float* codebooks,
-             int32_t nlist, int32_t M, int32_t dsub, int32_t d, float* S);
+             int32_t nlist, int32_t M, int32_t dsub, int32_t d, __half* S);
+void f32_to_f16(hipStream_t s, const float* in, int64_t n, __half* out);
 void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
                 const int64_t* csr_offsets, const uint8_t* csr_codes,
-                const float* S, const float* T, const float* coarse_dots,
+                const __half* S, const __half* T, const float* coarse_dots,
                 int32_t nlist, int32_t M, const int32_t* inv_offsets,
                 const int32_t* inv_q, const int32_t* inv_rank,
                 const int64_t* qp_off, const int64_t* q_cand_base,

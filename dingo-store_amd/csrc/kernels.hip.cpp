@@ -14,6 +14,7 @@
 //  * Plain library GEMMs (query x centroid / query x database dots) go
 //    through rocBLAS from the host side (dg_abi.cpp); everything irregular
 //    is hand-written here.
+#include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
@@ -717,7 +718,7 @@ __global__ void k_gather_codes(const uint8_t* __restrict__ src,
 __global__ void k_build_S(const float* __restrict__ centroids,
                           const float* __restrict__ codebooks, int32_t nlist,
                           int32_t M, int32_t dsub, int32_t d,
-                          float* __restrict__ S) {
+                          __half* __restrict__ S) {
   // one thread per (l, m, code): ||c_sub + cb||^2 over dsub elems
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t total = (int64_t)nlist * M * 256;
@@ -732,7 +733,13 @@ __global__ void k_build_S(const float* __restrict__ centroids,
     float t = c[j] + cb[j];
     acc += t * t;
   }
-  S[i] = acc;
+  S[i] = __float2half(acc);
+}
+
+__global__ void k_f32_to_f16(const float* __restrict__ in, int64_t n,
+                             __half* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = __float2half(in[i]);
 }
 
 // THE PQ scan: unit = (list, chunk of codes).  Codes for a 256-vector tile
@@ -742,8 +749,8 @@ __global__ void k_build_S(const float* __restrict__ centroids,
 // touch).  No cross-lane reduction.
 __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
-    const uint8_t* __restrict__ csr_codes, const float* __restrict__ S,
-    const float* __restrict__ T, const float* __restrict__ coarse_dots,
+    const uint8_t* __restrict__ csr_codes, const __half* __restrict__ S,
+    const __half* __restrict__ T, const float* __restrict__ coarse_dots,
     int32_t nlist, int32_t M, const int32_t* __restrict__ inv_offsets,
     const int32_t* __restrict__ inv_q, const int32_t* __restrict__ inv_rank,
     const int64_t* __restrict__ qp_off, const int64_t* __restrict__ q_cand_base,
@@ -760,7 +767,7 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
   const int32_t nql = inv_offsets[list + 1] - iq0;
   const int wave_id = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const float* Sl = S + (size_t)list * M * 256;
+  const __half* Sl = S + (size_t)list * M * 256;
 
   const int32_t TILE = 256;  // vectors staged per pass (TILE*M bytes LDS)
   for (int64_t t0 = row_start; t0 < row_end; t0 += TILE) {
@@ -783,7 +790,7 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
     for (int32_t qi = wave_id; qi < nql; qi += blockDim.x / WAVE) {
       const int32_t q = inv_q[iq0 + qi];
       const int32_t rank = inv_rank[iq0 + qi];
-      const float* Tq = T + (size_t)q * M * 256;
+      const __half* Tq = T + (size_t)q * M * 256;
       const float dot = coarse_dots[(size_t)q * nlist + list];
       const int64_t cb0 = q_cand_base[q] +
                           qp_off[(int64_t)q * nprobe + rank] - list_start;
@@ -797,18 +804,21 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
             (const uint32_t*)(lds_codes + (size_t)(lane + r * WAVE) * M);
       }
       if (metric == 0) {
-        const float* Sm = Sl;
-        const float* Tm = Tq;
+        const __half* Sm = Sl;
+        const __half* Tm = Tq;
         for (int32_t m4 = 0; m4 < M / 4; m4++) {
 #pragma unroll
           for (int r = 0; r < RPV; r++) {
             const uint32_t cw = code4[r][m4];
             const uint32_t c0 = cw & 255, c1 = (cw >> 8) & 255,
                            c2 = (cw >> 16) & 255, c3 = cw >> 24;
-            acc[r] +=
-                (Sm[c0] + Sm[256 + c1] + Sm[512 + c2] + Sm[768 + c3]) -
-                2.0f * (Tm[c0] + Tm[256 + c1] + Tm[512 + c2] +
-                        Tm[768 + c3]);
+            acc[r] += (__half2float(Sm[c0]) + __half2float(Sm[256 + c1]) +
+                       __half2float(Sm[512 + c2]) +
+                       __half2float(Sm[768 + c3])) -
+                      2.0f * (__half2float(Tm[c0]) +
+                              __half2float(Tm[256 + c1]) +
+                              __half2float(Tm[512 + c2]) +
+                              __half2float(Tm[768 + c3]));
           }
           Sm += 1024;
           Tm += 1024;
@@ -816,13 +826,15 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
 #pragma unroll
         for (int r = 0; r < RPV; r++) acc[r] -= 2.0f * dot;  // +qnorm@emit
       } else {
-        const float* Tm = Tq;
+        const __half* Tm = Tq;
         for (int32_t m4 = 0; m4 < M / 4; m4++) {
 #pragma unroll
           for (int r = 0; r < RPV; r++) {
             const uint32_t cw = code4[r][m4];
-            acc[r] += Tm[cw & 255] + Tm[256 + ((cw >> 8) & 255)] +
-                      Tm[512 + ((cw >> 16) & 255)] + Tm[768 + (cw >> 24)];
+            acc[r] += __half2float(Tm[cw & 255]) +
+                      __half2float(Tm[256 + ((cw >> 8) & 255)]) +
+                      __half2float(Tm[512 + ((cw >> 16) & 255)]) +
+                      __half2float(Tm[768 + (cw >> 24)]);
           }
           Tm += 1024;
         }
@@ -1434,15 +1446,20 @@ void gather_codes(hipStream_t s, const uint8_t* src, const uint32_t* perm,
 }
 
 void build_S(hipStream_t s, const float* centroids, const float* codebooks,
-             int32_t nlist, int32_t M, int32_t dsub, int32_t d, float* S) {
+             int32_t nlist, int32_t M, int32_t dsub, int32_t d, __half* S) {
   int64_t total = (int64_t)nlist * M * 256;
   hipLaunchKernelGGL(k_build_S, dim3(ceil_div(total, 256)), dim3(256), 0, s,
                      centroids, codebooks, nlist, M, dsub, d, S);
 }
 
+void f32_to_f16(hipStream_t s, const float* in, int64_t n, __half* out) {
+  if (n) hipLaunchKernelGGL(k_f32_to_f16, dim3(ceil_div(n, 256)), dim3(256),
+                            0, s, in, n, out);
+}
+
 void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
                 const int64_t* csr_offsets, const uint8_t* csr_codes,
-                const float* S, const float* T, const float* coarse_dots,
+                const __half* S, const __half* T, const float* coarse_dots,
                 int32_t nlist, int32_t M, const int32_t* inv_offsets,
                 const int32_t* inv_q, const int32_t* inv_rank,
                 const int64_t* qp_off, const int64_t* q_cand_base,
