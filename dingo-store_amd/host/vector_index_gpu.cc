@@ -317,5 +317,79 @@ extern "C" int dg_mirror_selftest(void) {
     for (auto& vd : res2[0].vector_with_distances)
       if (vd.vector_with_id.id >= 100) return 10;
   }
+
+  // ---- IVF lifecycle through the mirror (Train/Add/Search/RangeSearch/
+  // Upsert/Delete/Save/Load), L2 ----
+  {
+    const int32_t d2 = 32, n2 = 2000;
+    auto ivf = NewIvfFlatIndex(MetricType::kL2, d2, 16);
+    if (!ivf) return 20;
+    std::vector<VectorWithId> batch(n2);
+    uint32_t s = 99;
+    for (int i = 0; i < n2; i++) {
+      batch[i].id = i;
+      batch[i].vector.dimension = d2;
+      batch[i].vector.float_values.resize(d2);
+      for (int j = 0; j < d2; j++) {
+        s = s * 1664525u + 1013904223u;
+        batch[i].vector.float_values[j] = (s >> 8) * (1.0f / 16777216.0f);
+      }
+    }
+    if (ivf->IsTrained()) return 21;           // NeedTrain before Train
+    if (!ivf->NeedTrain()) return 22;
+    // untrained Search: blank results, OK (ivf_flat.cc:223-227)
+    std::vector<VectorWithDistanceResult> r0;
+    VectorSearchParameter p;
+    p.ivf_flat_nprobe = 16;
+    if (!ivf->Search({batch[0]}, 3, {}, false, p, r0).ok()) return 23;
+    if (!r0[0].vector_with_distances.empty()) return 24;
+    if (!ivf->Train(batch).ok()) return 25;
+    if (!ivf->IsTrained()) return 26;
+    if (!ivf->Add(batch).ok()) return 27;
+    int64_t cnt = 0;
+    ivf->GetCount(cnt);
+    if (cnt != n2) return 28;
+    std::vector<VectorWithDistanceResult> r1;
+    if (!ivf->Search({batch[7]}, 3, {}, false, p, r1).ok()) return 29;
+    if (r1[0].vector_with_distances.empty() ||
+        r1[0].vector_with_distances[0].vector_with_id.id != 7)
+      return 30;
+    // range search around the self-distance
+    std::vector<VectorWithDistanceResult> r2;
+    if (!ivf->RangeSearch({batch[7]}, 0.5f, {}, false, p, r2).ok())
+      return 31;
+    bool has_self = false;
+    for (auto& vd : r2[0].vector_with_distances)
+      if (vd.vector_with_id.id == 7) has_self = true;
+    if (!has_self) return 32;
+    // delete + upsert
+    if (!ivf->Delete({7}).ok()) return 33;
+    std::vector<VectorWithDistanceResult> r3;
+    ivf->Search({batch[7]}, 1, {}, false, p, r3);
+    if (!r3[0].vector_with_distances.empty() &&
+        r3[0].vector_with_distances[0].vector_with_id.id == 7)
+      return 34;
+    if (!ivf->Upsert({batch[7]}).ok()) return 35;
+    std::vector<VectorWithDistanceResult> r4;
+    ivf->Search({batch[7]}, 1, {}, false, p, r4);
+    if (r4[0].vector_with_distances.empty() ||
+        r4[0].vector_with_distances[0].vector_with_id.id != 7)
+      return 36;
+    // save / load round trip
+    const char* path = "/tmp/dg_mirror_selftest.dgi";
+    if (!ivf->Save(path).ok()) return 37;
+    if (!ivf->Load(path).ok()) return 38;
+    std::vector<VectorWithDistanceResult> r5;
+    ivf->Search({batch[7]}, 1, {}, false, p, r5);
+    if (r5[0].vector_with_distances.empty() ||
+        r5[0].vector_with_distances[0].vector_with_id.id != 7)
+      return 39;
+    // dimension mismatch rejected (CheckVectorDimension semantics)
+    VectorWithId bad;
+    bad.id = 999999;
+    bad.vector.dimension = d2 / 2;
+    bad.vector.float_values.resize(d2 / 2);
+    if (ivf->Add({bad}).ok()) return 40;
+  }
   return 0;
 }

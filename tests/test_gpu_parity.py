@@ -539,3 +539,46 @@ def test_concurrent_search():
             assert all(ex.map(worker, range(16)))
     finally:
         gpu.close()
+
+
+# ---------------- error paths (round-1 limits fail loudly) ----------------
+def test_error_paths():
+    with pytest.raises(dg.DgError):  # d % 4 != 0
+        dg.Index(dg.FLAT, dg.L2, 30)
+    with pytest.raises(dg.DgError):  # IVF d > 2048
+        dg.Index(dg.IVF_FLAT, dg.L2, 4096, nlist=16)
+    with pytest.raises(dg.DgError):  # PQ d % m != 0
+        dg.Index(dg.IVF_PQ, dg.L2, 64, nlist=16, m=7)
+    base, q = make_data(n=2000, d=32, nq=4)
+    idx = dg.Index(dg.FLAT, dg.L2, 32)
+    try:
+        idx.add(np.arange(2000, dtype=np.int64), base)
+        with pytest.raises(dg.DgError):  # k > 128 unsupported this round
+            idx.search(q, 200)
+    finally:
+        idx.close()
+    gpu, _ = build_pair(orc.L2, base, 16)
+    try:
+        gd, gi = gpu.search(q, 5, nprobe=0)  # default nprobe path
+        assert (gi[:, 0] >= 0).all()
+    finally:
+        gpu.close()
+
+
+def test_cosine_range_and_pq_ip():
+    """cosine range search + IP-metric PQ parity smoke."""
+    base, q = make_data(n=5000, d=32, nq=8)
+    idx = dg.Index(dg.FLAT, dg.COSINE, 32)
+    try:
+        idx.add(np.arange(5000, dtype=np.int64), base)
+        gd, _ = idx.search(q, 10)
+        radius = float(np.median(gd[:, 5]))  # faiss score threshold
+        lims, dists, ids = idx.range_search(q, radius)
+        assert (dists > radius).all()
+        ob = orc.normalize(base.copy())
+        oq = orc.normalize(q.copy())
+        ol, od, oi = orc.flat_range_search(orc.COSINE, ob, oq, radius)
+        assert np.array_equal(lims, ol)
+        assert np.array_equal(ids, oi)
+    finally:
+        idx.close()
