@@ -279,3 +279,32 @@ def test_golden_fixtures():
                             int(g["nprobe"]))
     assert np.array_equal(vi, g["ivf_ids"])
     assert np.array_equal(vd, g["ivf_dist"])
+
+
+def test_golden_pq_and_range():
+    g = np.load(os.path.join(GOLDEN, "oracle_golden.npz"))
+    base = workload.gen_base(int(g["seed"]), int(g["n"]), int(g["d"]))
+    q = workload.gen_queries(int(g["seed"]), int(g["n"]), int(g["d"]),
+                             int(g["nq"]))
+    nlist, k, nprobe = int(g["nlist"]), int(g["k"]), int(g["nprobe"])
+    cents = g["centroids"]
+    assign = orc.ivf_assign(orc.L2, base, cents)
+    off, gv, gi_ = orc.ivf_build(base, None, nlist, assign)
+    m = int(g["pq_m"])
+    cb = orc.pq_train(base - cents[assign], m)
+    assert np.array_equal(cb, g["codebooks"])  # deterministic PQ train
+    codes = orc.ivfpq_encode(base, assign, cents, cb)
+    gcodes = np.empty_like(codes)
+    cursor = off[:-1].copy()
+    for i in range(int(g["n"])):
+        gcodes[cursor[assign[i]]] = codes[i]
+        cursor[assign[i]] += 1
+    pd, pi = orc.ivfpq_search(orc.L2, cents, off, gcodes, gi_, cb, q, k,
+                              nprobe)
+    assert np.array_equal(pi, g["pq_ids"])
+    assert np.array_equal(pd, g["pq_dist"])
+    rl, rd, ri = orc.flat_range_search(orc.L2, base, q,
+                                       float(g["range_radius"]))
+    assert np.array_equal(rl, g["range_lims"])
+    assert np.array_equal(ri, g["range_ids"])
+    assert np.array_equal(rd, g["range_dist"])
