@@ -767,9 +767,22 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
   const int32_t nql = inv_offsets[list + 1] - iq0;
   const int wave_id = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const __half* Sl = S + (size_t)list * M * 256;
-
   const int32_t TILE = 256;  // vectors staged per pass (TILE*M bytes LDS)
+
+  // stage S_l into LDS once per unit (M*256 f16 = 48 KB at M=96); without
+  // this every (wave, query, m) visit re-fetches a 512 B row from L2/HBM -
+  // S exceeds L3 at cfg D and those fetches were the kernel's bound
+  __half* lds_S = (__half*)(lds_codes + (size_t)TILE * M);
+  {
+    const uint32_t* src_s = (const uint32_t*)(S + (size_t)list * M * 256);
+    uint32_t* dst_s = (uint32_t*)lds_S;
+    const int32_t words = M * 256 / 2;
+    for (int32_t w = threadIdx.x; w < words; w += blockDim.x)
+      dst_s[w] = src_s[w];
+  }
+  __syncthreads();
+  const __half* Sl = lds_S;
+
   for (int64_t t0 = row_start; t0 < row_end; t0 += TILE) {
     const int32_t tn = (int32_t)min((int64_t)TILE, row_end - t0);
     __syncthreads();
@@ -1466,7 +1479,7 @@ void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
                 int32_t nprobe, int metric, const uint32_t* bitmap,
                 int32_t chunk_rows, uint64_t* cand) {
   if (!n_units) return;
-  size_t lds = 256 * (size_t)M;
+  size_t lds = 256 * (size_t)M + (size_t)M * 256 * 2;  // codes + f16 S_l
   hipLaunchKernelGGL(k_ivfpq_scan, dim3((uint32_t)n_units), dim3(256), lds, s,
                      units, csr_offsets, csr_codes, S, T, coarse_dots, nlist,
                      M, inv_offsets, inv_q, inv_rank, qp_off, q_cand_base,
