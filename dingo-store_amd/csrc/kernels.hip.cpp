@@ -521,7 +521,7 @@ __global__ void k_tombstone(const int64_t* __restrict__ pos, int64_t n,
 // was bound by its 6-deep dependent shuffle chain per (row, query)).
 // Column loads are 16 B/lane coalesced; query element broadcasts come from
 // LDS.  Unused query slots are zero-staged so the inner loop is branch-free.
-template <int QTM, int RPL>  // queries per tile, rows per lane
+template <int QTM, int RPL, bool NT = false>  // queries/tile, rows/lane
 __device__ __forceinline__ void ivf_scan_col_body(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
@@ -604,7 +604,12 @@ __device__ __forceinline__ void ivf_scan_col_body(
           if (RPL >= 4) {
 #pragma unroll
             for (int v = 0; v < RPL / 4; v++) {
-              const float4 c4 = ((const float4*)cp)[v];
+              // nt (stream-once) column loads selectable: each chunk is
+              // read by exactly one CU per batch, so L1/L2 retention buys
+              // nothing (MI355X_MICROARCH.md nt-weights row)
+              const float4 c4 = NT
+                  ? __builtin_nontemporal_load((const float4*)cp + v)
+                  : ((const float4*)cp)[v];
               c[u][4 * v + 0] = c4.x;
               c[u][4 * v + 1] = c4.y;
               c[u][4 * v + 2] = c4.z;
@@ -1041,6 +1046,23 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_col(
 }
 
 template <int QTM, int RPL>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_col_nt(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  ivf_scan_col_body<QTM, RPL, true>(units, csr_offsets, chunk_off, chunk_base,
+                                    tvec, vnorms, queries, d, inv_offsets,
+                                    inv_q, inv_rank, qp_off, q_cand_base,
+                                    nprobe, metric, bitmap, chunk_rows, cand);
+}
+
+template <int QTM, int RPL>
 __global__ void __launch_bounds__(256, 3) k_ivf_scan_col_hi(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
@@ -1370,6 +1392,17 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
     case 3: DG_SCAN_LAUNCH(8, 8); break;
     case 4: DG_SCAN_LAUNCH(4, 4); break;
     case 5: DG_SCAN_LAUNCH(12, 4); break;
+    case 8: {
+      constexpr int QTM = 16;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_col_nt<QTM, 4>),
+                         dim3((uint32_t)n_units), dim3(256), lds, s, units,
+                         csr_offsets, chunk_off, chunk_base, tvec, vnorms,
+                         queries, d, inv_offsets, inv_q, inv_rank, qp_off,
+                         q_cand_base, nprobe, metric, bitmap, chunk_rows,
+                         cand);
+      break;
+    }
     case 7: {
       constexpr int QTM = 16, DEPTH = 6;
       size_t lds = (size_t)QTM * d * 4 + 4 * DEPTH * 4 * 256 * 4 + QTM * 8;
