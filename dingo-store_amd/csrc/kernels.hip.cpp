@@ -607,9 +607,15 @@ __device__ __forceinline__ void ivf_scan_col_body(
               // nt (stream-once) column loads selectable: each chunk is
               // read by exactly one CU per batch, so L1/L2 retention buys
               // nothing (MI355X_MICROARCH.md nt-weights row)
-              const float4 c4 = NT
-                  ? __builtin_nontemporal_load((const float4*)cp + v)
-                  : ((const float4*)cp)[v];
+              typedef float nf4 __attribute__((ext_vector_type(4)));
+              float4 c4;
+              if (NT) {
+                const nf4 nv =
+                    __builtin_nontemporal_load((const nf4*)cp + v);
+                c4 = make_float4(nv[0], nv[1], nv[2], nv[3]);
+              } else {
+                c4 = ((const float4*)cp)[v];
+              }
               c[u][4 * v + 0] = c4.x;
               c[u][4 * v + 1] = c4.y;
               c[u][4 * v + 2] = c4.z;
