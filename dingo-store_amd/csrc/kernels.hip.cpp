@@ -753,11 +753,12 @@ __global__ void k_f32_to_f16(const float* __restrict__ in, int64_t n,
   if (i < n) out[i] = __float2half(in[i]);
 }
 
-// THE PQ scan: unit = (list, chunk of codes).  Codes for a 256-vector tile
+// THE PQ scan: unit = (list, chunk of codes).  Codes for a RPV*64-vector tile
 // are LDS-staged once and reused by every probing query; each LANE owns one
 // vector and walks m = 0..M-1 sequentially, so all lanes of a wave gather
 // within the same 1 KB rows of T[q] and S[l] (L1-resident after first
 // touch).  No cross-lane reduction.
+template <int RPV>  // rows (vectors) per lane; TILE = RPV * 64
 __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const uint8_t* __restrict__ csr_codes, const __half* __restrict__ S,
@@ -778,7 +779,7 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
   const int32_t nql = inv_offsets[list + 1] - iq0;
   const int wave_id = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int32_t TILE = 256;  // vectors staged per pass (TILE*M bytes LDS)
+  const int32_t TILE = RPV * WAVE;  // vectors staged per pass
 
   // stage S_l into LDS once per unit (M*256 f16 = 48 KB at M=96); without
   // this every (wave, query, m) visit re-fetches a 512 B row from L2/HBM -
@@ -818,7 +819,6 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
       const float dot = coarse_dots[(size_t)q * nlist + list];
       const int64_t cb0 = q_cand_base[q] +
                           qp_off[(int64_t)q * nprobe + rank] - list_start;
-      constexpr int RPV = 4;  // TILE = RPV * WAVE
       float acc[RPV];
       const uint32_t* code4[RPV];
 #pragma unroll
@@ -1518,11 +1518,25 @@ void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
                 int32_t nprobe, int metric, const uint32_t* bitmap,
                 int32_t chunk_rows, uint64_t* cand) {
   if (!n_units) return;
-  size_t lds = 256 * (size_t)M + (size_t)M * 256 * 2;  // codes + f16 S_l
-  hipLaunchKernelGGL(k_ivfpq_scan, dim3((uint32_t)n_units), dim3(256), lds, s,
-                     units, csr_offsets, csr_codes, S, T, coarse_dots, nlist,
-                     M, inv_offsets, inv_q, inv_rank, qp_off, q_cand_base,
-                     nprobe, metric, bitmap, chunk_rows, cand);
+  static int rpv = []() {
+    const char* e = getenv("DG_PQ_RPV");
+    return e ? atoi(e) : 4;
+  }();
+  if (rpv == 8) {  // TILE=512: halves per-tile T-row re-fetches, 1 block/CU
+    size_t lds = 512 * (size_t)M + (size_t)M * 256 * 2;
+    hipLaunchKernelGGL((k_ivfpq_scan<8>), dim3((uint32_t)n_units), dim3(256),
+                       lds, s, units, csr_offsets, csr_codes, S, T,
+                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
+                       qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
+  } else {
+    size_t lds = 256 * (size_t)M + (size_t)M * 256 * 2;  // codes + f16 S_l
+    hipLaunchKernelGGL((k_ivfpq_scan<4>), dim3((uint32_t)n_units), dim3(256),
+                       lds, s, units, csr_offsets, csr_codes, S, T,
+                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
+                       qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
+  }
 }
 
 void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
