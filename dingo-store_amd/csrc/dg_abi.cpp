@@ -1821,7 +1821,9 @@ extern "C" dg_status dg_stats(dg_index* ix, dg_stats_out* out) {
   size_t db = 0;
   for (auto b : {&ix->d_vectors, &ix->d_ids, &ix->d_assign, &ix->d_centroids,
                  &ix->d_cnorms, &ix->d_csr_offsets, &ix->d_csr_vectors,
-                 &ix->d_csr_ids, &ix->d_csr_vnorms})
+                 &ix->d_csr_ids, &ix->d_csr_vnorms, &ix->d_csr_t,
+                 &ix->d_codebooks, &ix->d_codes, &ix->d_csr_codes, &ix->d_S,
+                 &ix->d_cb_norms})
     db += b->cap;
   out->device_bytes = (int64_t)db;
   if (ix->times.last_nq > 0) {

@@ -63,7 +63,7 @@ Status pack(const std::vector<VectorWithId>& v, int32_t dim,
 class GpuIndexBase : public VectorIndex {
  public:
   GpuIndexBase(dg_index_kind kind, MetricType metric, int32_t dim,
-               int32_t nlist, int device) {
+               int32_t nlist, int device, int32_t pq_m = 0) {
     metric_ = metric;
     dim_ = dim;
     dg_index_desc desc{};
@@ -71,6 +71,8 @@ class GpuIndexBase : public VectorIndex {
     desc.metric = (int32_t)metric;
     desc.d = dim;
     desc.nlist = nlist;
+    desc.pq_m = pq_m;
+    desc.pq_nbits = 8;
     desc.device = device;
     create_st_ = dg_index_create(&idx_, &desc);
   }
@@ -254,6 +256,14 @@ class GpuIvfFlatIndex : public GpuIndexBase {
   bool NeedTrain() override { return !IsTrained(); }
 };
 
+class GpuIvfPqIndex : public GpuIndexBase {
+ public:
+  GpuIvfPqIndex(MetricType m, int32_t d, int32_t nlist, int32_t pq_m,
+                int dev)
+      : GpuIndexBase(DG_INDEX_IVF_PQ, m, d, nlist, dev, pq_m) {}
+  bool NeedTrain() override { return !IsTrained(); }
+};
+
 }  // namespace
 
 std::unique_ptr<VectorIndex> NewFlatIndex(MetricType metric, int32_t dim,
@@ -266,6 +276,15 @@ std::unique_ptr<VectorIndex> NewFlatIndex(MetricType metric, int32_t dim,
 std::unique_ptr<VectorIndex> NewIvfFlatIndex(MetricType metric, int32_t dim,
                                              int32_t ncentroids, int device) {
   auto p = std::make_unique<GpuIvfFlatIndex>(metric, dim, ncentroids, device);
+  if (!p->CreateStatus().ok()) return nullptr;
+  return p;
+}
+
+std::unique_ptr<VectorIndex> NewIvfPqIndex(MetricType metric, int32_t dim,
+                                           int32_t ncentroids,
+                                           int32_t nsubvector, int device) {
+  auto p = std::make_unique<GpuIvfPqIndex>(metric, dim, ncentroids,
+                                           nsubvector, device);
   if (!p->CreateStatus().ok()) return nullptr;
   return p;
 }

@@ -152,6 +152,10 @@ std::unique_ptr<VectorIndex> NewFlatIndex(MetricType metric, int32_t dim,
 std::unique_ptr<VectorIndex> NewIvfFlatIndex(MetricType metric, int32_t dim,
                                              int32_t ncentroids,
                                              int device = -1);
+std::unique_ptr<VectorIndex> NewIvfPqIndex(MetricType metric, int32_t dim,
+                                           int32_t ncentroids,
+                                           int32_t nsubvector,
+                                           int device = -1);
 
 }  // namespace dingogpu
 

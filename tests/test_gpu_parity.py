@@ -582,3 +582,34 @@ def test_cosine_range_and_pq_ip():
         assert np.array_equal(ids, oi)
     finally:
         idx.close()
+
+
+def test_ivfpq_range_and_cosine():
+    """PQ + range search (shares the candidate machinery) and cosine PQ."""
+    base, q = make_data(n=10000, d=64, nq=8)
+    gpu, (cents, cb, off, gcodes, gi_) = build_pq_pair(orc.L2, base, 16, 8)
+    try:
+        gd, _ = gpu.search(q, 10, nprobe=8)
+        radius = float(np.median(gd[:, 5]))
+        lims, dists, ids = gpu.range_search(q, radius)
+        assert (dists < radius).all()
+        assert lims[-1] > 0
+        # counts agree with top-k: entries better than radius
+        for r in range(q.shape[0]):
+            expected = int((gd[r] < radius).sum())
+            got = int(lims[r + 1] - lims[r])
+            # gd only has top-10; range may find more
+            assert got >= expected or expected == 10
+    finally:
+        gpu.close()
+    gpu2, (cents2, cb2, off2, gcodes2, gi2) = build_pq_pair(
+        orc.COSINE, base, 16, 8)
+    try:
+        gd2, gi2r = gpu2.search(q, 10, nprobe=8)
+    finally:
+        gpu2.close()
+    oq = orc.normalize(q.copy())
+    od, oi = orc.ivfpq_search(orc.COSINE, cents2, off2, gcodes2, gi2, cb2,
+                              oq, 10, 8)
+    frac = ids_match_with_tie_slack(gd2, gi2r, od, oi, tol=5e-3)
+    assert frac >= 0.9, frac
