@@ -129,10 +129,24 @@ struct DeviceGuard {
   }
 };
 
-// sgemm: dots[rows x cols] (row-major) = X[rows x d] * Y[cols x d]^T
+// dots[rows x cols] (row-major) = X[rows x d] * Y[cols x d]^T.
+// The hand f32 MFMA kernel (k_dots_mfma — the north star's coarse-assign
+// MFMA target) runs when the shape fills its 64x64 tiles; rocBLAS sgemm
+// covers small/odd shapes and DG_GEMM=rocblas forces it for A/B.
 static dg_status sgemm_dots(dg_index* ix, const float* X, int64_t rows,
                             const float* Y, int64_t cols, int32_t d,
                             float* dots) {
+  static int force = []() {
+    const char* e = getenv("DG_GEMM");
+    return e ? (strcmp(e, "rocblas") == 0 ? 1 : 2) : 0;
+  }();
+  const bool use_mfma =
+      force != 1 &&
+      (force == 2 || (rows >= 48 && cols >= 64 && d >= 64));
+  if (use_mfma) {
+    dgk::dots_mfma(ix->stream, X, rows, Y, cols, d, dots, cols);
+    return DG_OK;
+  }
   const float one = 1.0f, zero = 0.0f;
   DG_ROCBLAS_CHECK(rocblas_sgemm(
       ix->blas, rocblas_operation_transpose, rocblas_operation_none,

@@ -239,6 +239,8 @@ void emit_results(hipStream_t s, const uint64_t* topk,
                   int32_t k, int metric, int add_qnorm, float* out_dist,
                   int64_t* out_ids);
 // k-means / finalize helpers
+void dots_mfma(hipStream_t s, const float* X, int64_t M, const float* Y,
+               int64_t N, int32_t K, float* C, int64_t ldc);
 void hist_assign(hipStream_t s, const int32_t* assign, int64_t n,
                  int32_t nlist, int32_t* counts);
 void scatter_perm(hipStream_t s, const int32_t* assign, int64_t n,

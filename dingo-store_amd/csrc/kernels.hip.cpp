@@ -1126,6 +1126,81 @@ __global__ void k_transpose_chunks(
   }
 }
 
+// ---------- hand MFMA f32 dots GEMM ----------
+// C[M][N] = X[M][d] . Y[N][d]^T, all row-major — the coarse
+// query x centroid assignment GEMM the north star names as THE MFMA target
+// (SURVEY.md §8a row a4), also used for the Flat exact-scan dots.
+// gfx950 f32-input MFMA v_mfma_f32_32x32x2_f32: exact f32 (a k-ordered fmaf
+// chain, cdna_hip_programming.md §3), 64 cyc issue = dependent latency, so
+// one accumulator per wave reaches the 157 TF f32 rate.  Block = 4 waves =
+// 64x64 tile (each wave one 32x32 accumulator, 16 AGPRs), BK=16 double-
+// buffered LDS (2 x 2 x 64x16 f32 = 16 KB), A/B operands read from LDS as
+// b64 (two k-steps per read).
+__global__ void __launch_bounds__(256, 4) k_dots_mfma(
+    const float* __restrict__ X, const float* __restrict__ Y, int64_t M,
+    int64_t N, int32_t K, float* __restrict__ C, int64_t ldc) {
+  constexpr int BM = 64, BN = 64, BK = 16;
+  __shared__ float lx[2][BM][BK + 1];  // +1: column reads conflict-free
+  __shared__ float ly[2][BN][BK + 1];
+  using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
+
+  const int64_t m0 = (int64_t)blockIdx.y * BM;
+  const int64_t n0 = (int64_t)blockIdx.x * BN;
+  const int wave_id = threadIdx.x / WAVE;  // 2x2 wave grid of 32x32 tiles
+  const int lane = threadIdx.x % WAVE;
+  const int wm = (wave_id >> 1) * 32;  // wave row offset in tile
+  const int wn = (wave_id & 1) * 32;   // wave col offset
+
+  // cooperative staging: 256 threads load 64x16 = 1024 floats per operand
+  // (4 per thread); global layout row-major [64][K] -> coalesced along K
+  const int sr = threadIdx.x / 4;        // staged row 0..63
+  const int sk = (threadIdx.x % 4) * 4;  // staged k 0..12 step 4
+
+  f32x16 acc = {};  // one 32x32 f32 accumulator (16 VGPRs)
+  auto stage = [&](int buf, int32_t k0) {
+    const int64_t xm = m0 + sr;
+    const int64_t yn = n0 + sr;
+#pragma unroll
+    for (int t = 0; t < 4; t++) {
+      const int32_t k = k0 + sk + t;
+      lx[buf][sr][sk + t] =
+          (xm < M && k < K) ? X[xm * K + k] : 0.f;
+      ly[buf][sr][sk + t] =
+          (yn < N && k < K) ? Y[yn * K + k] : 0.f;
+    }
+  };
+  auto compute = [&](int buf) {
+    const int ar = wm + (lane & 31);   // A row this lane feeds
+    const int br = wn + (lane & 31);   // B col this lane feeds
+    const int kh = (lane >> 5);        // k half (0/1) within each pair
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 2) {
+      const float a = lx[buf][ar][kk + kh];
+      const float b = ly[buf][br][kk + kh];
+      acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  int buf = 0;
+  for (int32_t k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    if (k0 + BK < K) stage(buf ^ 1, k0 + BK);
+    compute(buf);
+    buf ^= 1;
+    __syncthreads();
+  }
+  // C/D layout for 32x32 f32 MFMA: col = lane&31,
+  // row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)   (cdna_hip_programming.md §3)
+#pragma unroll
+  for (int reg = 0; reg < 16; reg++) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+    const int64_t cm = m0 + wm + row;
+    const int64_t cn = n0 + wn + (lane & 31);
+    if (cm < M && cn < N) C[cm * ldc + cn] = acc[reg];
+  }
+}
+
 // ---------- k-means / finalize helpers ----------
 __global__ void k_hist_assign(const int32_t* __restrict__ assign, int64_t n,
                               int32_t* __restrict__ counts) {
@@ -1547,6 +1622,13 @@ void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
   hipLaunchKernelGGL(k_transpose_chunks, dim3((uint32_t)n_units), dim3(256),
                      0, s, units, csr_offsets, chunk_off, chunk_base,
                      rowmajor, d, chunk_rows, tvec);
+}
+
+void dots_mfma(hipStream_t s, const float* X, int64_t M, const float* Y,
+               int64_t N, int32_t K, float* C, int64_t ldc) {
+  dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64));
+  hipLaunchKernelGGL(k_dots_mfma, grid, dim3(256), 0, s, X, Y, M, N, K, C,
+                     ldc);
 }
 
 void hist_assign(hipStream_t s, const int32_t* assign, int64_t n,
