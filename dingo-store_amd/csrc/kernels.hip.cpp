@@ -1136,48 +1136,74 @@ __global__ void k_transpose_chunks(
 // 64x64 tile (each wave one 32x32 accumulator, 16 AGPRs), BK=16 double-
 // buffered LDS (2 x 2 x 64x16 f32 = 16 KB), A/B operands read from LDS as
 // b64 (two k-steps per read).
-__global__ void __launch_bounds__(256, 4) k_dots_mfma(
+__global__ void __launch_bounds__(256, 2) k_dots_mfma(
     const float* __restrict__ X, const float* __restrict__ Y, int64_t M,
     int64_t N, int32_t K, float* __restrict__ C, int64_t ldc) {
-  constexpr int BM = 64, BN = 64, BK = 16;
-  __shared__ float lx[2][BM][BK + 1];  // +1: column reads conflict-free
-  __shared__ float ly[2][BN][BK + 1];
+  // 128x128 block, 4 waves as 2x2 of 64x64, each wave 2x2 accumulators of
+  // 32x32 => 64 MFMA per wave between barriers (the guide's untuned
+  // 128x128x32 f32 shape reaches 122 TF; a 64x64/BK16 first cut measured
+  // 27 TF — barrier overhead per 8 MFMAs dominated).
+  constexpr int BM = 128, BN = 128, BK = 32;
+  __shared__ float lx[2][BM][BK + 1];  // odd stride: conflict-free column
+  __shared__ float ly[2][BN][BK + 1];  //   reads; scalar b32 staging writes
   using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
 
   const int64_t m0 = (int64_t)blockIdx.y * BM;
   const int64_t n0 = (int64_t)blockIdx.x * BN;
-  const int wave_id = threadIdx.x / WAVE;  // 2x2 wave grid of 32x32 tiles
+  const int wave_id = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int wm = (wave_id >> 1) * 32;  // wave row offset in tile
-  const int wn = (wave_id & 1) * 32;   // wave col offset
+  const int wm = (wave_id >> 1) * 64;  // wave row offset in block tile
+  const int wn = (wave_id & 1) * 64;   // wave col offset
 
-  // cooperative staging: 256 threads load 64x16 = 1024 floats per operand
-  // (4 per thread); global layout row-major [64][K] -> coalesced along K
-  const int sr = threadIdx.x / 4;        // staged row 0..63
-  const int sk = (threadIdx.x % 4) * 4;  // staged k 0..12 step 4
+  // staging: 128 rows x 32 k per operand; thread t loads 16 consecutive k
+  // of one row (4 x float4 when fully in range)
+  const int sr = threadIdx.x / 2;         // staged row 0..127
+  const int sk = (threadIdx.x % 2) * 16;  // staged k 0 or 16
 
-  f32x16 acc = {};  // one 32x32 f32 accumulator (16 VGPRs)
+  f32x16 acc[2][2] = {};
   auto stage = [&](int buf, int32_t k0) {
     const int64_t xm = m0 + sr;
     const int64_t yn = n0 + sr;
+    if (xm < M && yn < N && k0 + sk + 16 <= K) {
 #pragma unroll
-    for (int t = 0; t < 4; t++) {
-      const int32_t k = k0 + sk + t;
-      lx[buf][sr][sk + t] =
-          (xm < M && k < K) ? X[xm * K + k] : 0.f;
-      ly[buf][sr][sk + t] =
-          (yn < N && k < K) ? Y[yn * K + k] : 0.f;
+      for (int t = 0; t < 4; t++) {
+        const float4 vx = *(const float4*)&X[xm * K + k0 + sk + 4 * t];
+        const float4 vy = *(const float4*)&Y[yn * K + k0 + sk + 4 * t];
+        lx[buf][sr][sk + 4 * t + 0] = vx.x;
+        lx[buf][sr][sk + 4 * t + 1] = vx.y;
+        lx[buf][sr][sk + 4 * t + 2] = vx.z;
+        lx[buf][sr][sk + 4 * t + 3] = vx.w;
+        ly[buf][sr][sk + 4 * t + 0] = vy.x;
+        ly[buf][sr][sk + 4 * t + 1] = vy.y;
+        ly[buf][sr][sk + 4 * t + 2] = vy.z;
+        ly[buf][sr][sk + 4 * t + 3] = vy.w;
+      }
+    } else {  // edge tile: element-wise with zero padding
+#pragma unroll
+      for (int t = 0; t < 16; t++) {
+        const int32_t k = k0 + sk + t;
+        lx[buf][sr][sk + t] = (xm < M && k < K) ? X[xm * K + k] : 0.f;
+        ly[buf][sr][sk + t] = (yn < N && k < K) ? Y[yn * K + k] : 0.f;
+      }
     }
   };
   auto compute = [&](int buf) {
-    const int ar = wm + (lane & 31);   // A row this lane feeds
-    const int br = wn + (lane & 31);   // B col this lane feeds
-    const int kh = (lane >> 5);        // k half (0/1) within each pair
+    const int al = lane & 31;   // row/col within a 32-wide fragment
+    const int kh = (lane >> 5); // k half within each k-pair
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
-      const float a = lx[buf][ar][kk + kh];
-      const float b = ly[buf][br][kk + kh];
-      acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+      float a[2], b[2];
+#pragma unroll
+      for (int i = 0; i < 2; i++) {
+        a[i] = lx[buf][wm + 32 * i + al][kk + kh];
+        b[i] = ly[buf][wn + 32 * i + al][kk + kh];
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; mi++)
+#pragma unroll
+        for (int ni = 0; ni < 2; ni++)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+              a[mi], b[ni], acc[mi][ni], 0, 0, 0);
     }
   };
 
@@ -1193,12 +1219,16 @@ __global__ void __launch_bounds__(256, 4) k_dots_mfma(
   // C/D layout for 32x32 f32 MFMA: col = lane&31,
   // row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)   (cdna_hip_programming.md §3)
 #pragma unroll
-  for (int reg = 0; reg < 16; reg++) {
-    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
-    const int64_t cm = m0 + wm + row;
-    const int64_t cn = n0 + wn + (lane & 31);
-    if (cm < M && cn < N) C[cm * ldc + cn] = acc[reg];
-  }
+  for (int mi = 0; mi < 2; mi++)
+#pragma unroll
+    for (int ni = 0; ni < 2; ni++)
+#pragma unroll
+      for (int reg = 0; reg < 16; reg++) {
+        const int row = (reg & 3) + 8 * (reg >> 2) + 4 * (lane >> 5);
+        const int64_t cm = m0 + wm + 32 * mi + row;
+        const int64_t cn = n0 + wn + 32 * ni + (lane & 31);
+        if (cm < M && cn < N) C[cm * ldc + cn] = acc[mi][ni][reg];
+      }
 }
 
 // ---------- k-means / finalize helpers ----------
@@ -1626,7 +1656,7 @@ void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
 
 void dots_mfma(hipStream_t s, const float* X, int64_t M, const float* Y,
                int64_t N, int32_t K, float* C, int64_t ldc) {
-  dim3 grid((uint32_t)((N + 63) / 64), (uint32_t)((M + 63) / 64));
+  dim3 grid((uint32_t)((N + 127) / 128), (uint32_t)((M + 127) / 128));
   hipLaunchKernelGGL(k_dots_mfma, grid, dim3(256), 0, s, X, Y, M, N, K, C,
                      ldc);
 }
