@@ -215,6 +215,7 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
   rocblas_set_stream(ix->blas, ix->stream);
   for (auto& e : ix->ev) (void)hipEventCreate(&e);
   ix->events_ready = true;
+  (void)hipHostMalloc((void**)&ix->h_pinned, 64);
   if (desc.kind == DG_INDEX_FLAT) ix->trained = true;
   if (desc.reserve > 0) {  // capacity hint: avoids grow-copies during add
     // PQ never stores raw vectors (codes only); others reserve the vector
@@ -254,6 +255,7 @@ extern "C" void dg_index_destroy(dg_index* ix) {
     dbuf_free(*b);
   for (auto& e : ix->ev)
     if (e) (void)hipEventDestroy(e);
+  if (ix->h_pinned) (void)hipHostFree(ix->h_pinned);
   if (ix->blas) rocblas_destroy_handle(ix->blas);
   if (ix->stream) (void)hipStreamDestroy(ix->stream);
   delete ix;
@@ -918,8 +920,19 @@ static dg_status finalize_csr(dg_index* ix) {
                               ix->stream));
   DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
 
-  if (is_ivf && n > 0) {
-    // chunk geometry (host) + transpose into [d][nrows_pad] chunk layout
+  if ((is_ivf || is_pq) && n > 0) {
+    // candidate-buffer upper bounds: prefix of list lens sorted descending
+    // (lets searches size ws_cand without a device readback)
+    std::vector<int64_t> lens(nlist);
+    for (int32_t l = 0; l < nlist; l++)
+      lens[l] = ix->h_csr_offsets[l + 1] - ix->h_csr_offsets[l];
+    std::sort(lens.begin(), lens.end(), std::greater<int64_t>());
+    ix->h_len_prefix_desc.assign(nlist + 1, 0);
+    for (int32_t l = 0; l < nlist; l++)
+      ix->h_len_prefix_desc[l + 1] = ix->h_len_prefix_desc[l] + lens[l];
+
+    // chunk geometry (host; both IVF kinds use the all-chunks unit array;
+    // the transposed layout below is IVF-Flat only)
     std::vector<int32_t> chunk_off(nlist + 1, 0);
     std::vector<int64_t> chunk_base;
     std::vector<uint32_t> all_units;
@@ -942,7 +955,11 @@ static dg_status finalize_csr(dg_index* ix) {
                         (size_t)ix->total_chunks * 8 +
                         (size_t)ix->total_chunks * 8;
     if ((st = dbuf_reserve(ix->d_chunk_meta, meta_bytes, ix->stream,
-                           false)) != DG_OK ||
+                           false)) != DG_OK) {
+      dbuf_free(rm_tmp);
+      return st;
+    }
+    if (is_ivf &&
         // +1 chunk of slack: the glds scan's tail quarters issue padded
         // reads past the last chunk (results discarded at emit)
         (st = dbuf_reserve(ix->d_csr_t,
@@ -966,10 +983,11 @@ static dg_status finalize_csr(dg_index* ix) {
     (void)hipMemcpyAsync(d_all_units, all_units.data(),
                          (size_t)ix->total_chunks * 8, hipMemcpyHostToDevice,
                          ix->stream);
-    dgk::transpose_chunks(ix->stream, d_all_units, ix->total_chunks,
-                          d_offsets, d_chunk_off, d_chunk_base,
-                          (const float*)rowmajor.p, d, CR,
-                          (float*)ix->d_csr_t.p);
+    if (is_ivf)
+      dgk::transpose_chunks(ix->stream, d_all_units, ix->total_chunks,
+                            d_offsets, d_chunk_off, d_chunk_base,
+                            (const float*)rowmajor.p, d, CR,
+                            (float*)ix->d_csr_t.p);
     DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
   }
   dbuf_free(rm_tmp);
@@ -1362,60 +1380,43 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     dgk::cand_offsets(ix->stream, probes, nq, np,
                       (const int64_t*)ix->d_csr_offsets.p, qp_off, q_total);
     dgk::excl_scan_i64(ix->stream, q_total, nq, q_cand_base);
-    // units
     const int32_t chunk_rows = dg_index::kChunkRows;
-    dgk::fill_unit_counts(ix->stream, inv_counts, nlist,
-                          (const int64_t*)ix->d_csr_offsets.p, chunk_rows,
-                          unit_counts);
-    // reuse cursors space scan: unit offsets (i64 in ws, reuse inv_offsets?
-    // inv_offsets64 still needed by scan kernel; allocate unit offsets in
-    // ws_units along with the unit array)
-    // sizing sync: total candidates + total units + inv_counts (for stats)
-    dg_dbuf& wsu = ix->ws_units;
-    if ((st = dbuf_reserve(wsu, ((size_t)nlist + 1) * 8, ix->stream, false))
-        != DG_OK) {
+    // candidate buffer sized from the HOST upper bound (nq x sum of the
+    // np longest lists, finalize's sorted prefix) and the scan launched
+    // over ALL chunks (kernel exits in ~10 instructions for unprobed
+    // lists) — no mid-search device readback / stream sync.
+    const int64_t ub_cand =
+        nq * ix->h_len_prefix_desc[std::min<int32_t>(np, nlist)];
+    if ((st = dbuf_reserve(ix->ws_cand,
+                           (size_t)std::max<int64_t>(ub_cand, 1) * 8,
+                           ix->stream, false)) != DG_OK) {
       dbuf_free(d_fids);
       dbuf_free(ws_bitmap);
       return st;
     }
-    int64_t* unit_offsets = (int64_t*)wsu.p;
-    dgk::excl_scan_i32_to_i64(ix->stream, unit_counts, nlist, unit_offsets);
-    int64_t totals[2] = {0, 0};
-    std::vector<int32_t> h_inv_counts(nlist);
-    DG_HIP_CHECK(hipMemcpyAsync(&totals[0], q_cand_base + nq, 8,
-                                hipMemcpyDeviceToHost, ix->stream));
-    DG_HIP_CHECK(hipMemcpyAsync(&totals[1], unit_offsets + nlist, 8,
-                                hipMemcpyDeviceToHost, ix->stream));
-    DG_HIP_CHECK(hipMemcpyAsync(h_inv_counts.data(), inv_counts,
-                                (size_t)nlist * 4, hipMemcpyDeviceToHost,
-                                ix->stream));
-    DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
-    int64_t total_cand = totals[0];
-    int32_t total_units = (int32_t)totals[1];
-    // algorithmic bytes: one read of each probed list (IVF-Flat: vectors +
-    // norms; PQ: codes — the S/T table gathers are on-chip cache traffic,
-    // DESIGN.md §ivf-pq)
-    int64_t alg = 0;
+    char* cm = (char*)ix->d_chunk_meta.p;
+    const uint32_t* units =
+        (const uint32_t*)(cm + ((size_t)nlist + 1) * 4 +
+                          (size_t)ix->total_chunks * 8);
+    const int32_t total_units = ix->total_chunks;
+    // algorithmic bytes (roofline) summed on device, read back async into
+    // pinned memory; dg_stats synchronizes on ev[5]
     const int64_t row_bytes = is_pq ? ix->desc.pq_m : (int64_t)(d * 4 + 4);
-    for (int32_t l = 0; l < nlist; l++)
-      if (h_inv_counts[l] > 0)
-        alg += (ix->h_csr_offsets[l + 1] - ix->h_csr_offsets[l]) * row_bytes;
-    ix->times.last_scan_bytes_alg = alg;
-
-    size_t need = ((size_t)nlist + 1) * 8 + (size_t)total_units * 8;
-    if ((st = dbuf_reserve(wsu, need, ix->stream, true)) != DG_OK ||
-        (st = dbuf_reserve(ix->ws_cand, (size_t)total_cand * 8, ix->stream,
-                           false)) != DG_OK) {
-      dbuf_free(d_fids);
-      dbuf_free(ws_bitmap);
-      return st;
+    if (ix->h_pinned) {
+      dg_dbuf& wsu = ix->ws_units;
+      if ((st = dbuf_reserve(wsu, 8, ix->stream, false)) != DG_OK) {
+        dbuf_free(d_fids);
+        dbuf_free(ws_bitmap);
+        return st;
+      }
+      (void)hipMemsetAsync(wsu.p, 0, 8, ix->stream);
+      dgk::alg_bytes(ix->stream, inv_counts,
+                     (const int64_t*)ix->d_csr_offsets.p, nlist, row_bytes,
+                     (int64_t*)wsu.p);
+      (void)hipMemcpyAsync(ix->h_pinned, wsu.p, 8, hipMemcpyDeviceToHost,
+                           ix->stream);
+      (void)hipEventRecord(ix->ev[5], ix->stream);
     }
-    unit_offsets = (int64_t*)wsu.p;
-    uint32_t* units = (uint32_t*)(unit_offsets + nlist + 1);
-    // fill units from unit_offsets (i64 -> i32 via cursors copy)
-    dgk::init_cursors(ix->stream, unit_offsets, nlist, cursors);
-    dgk::fill_units(ix->stream, cursors, unit_counts, nlist, nullptr, 0,
-                    units, total_units);
 
     (void)hipEventRecord(ix->ev[2], ix->stream);
     if (is_pq) {
@@ -1845,6 +1846,10 @@ extern "C" dg_status dg_stats(dg_index* ix, dg_stats_out* out) {
   out->device_bytes = (int64_t)db;
   if (ix->times.last_nq > 0) {
     (void)hipEventSynchronize(ix->ev[4]);
+    if (ix->h_pinned && ix->desc.kind != DG_INDEX_FLAT) {
+      (void)hipEventSynchronize(ix->ev[5]);
+      ix->times.last_scan_bytes_alg = *ix->h_pinned;
+    }
     float ms01 = 0, ms23 = 0, ms04 = 0;
     (void)hipEventElapsedTime(&ms01, ix->ev[0], ix->ev[1]);
     (void)hipEventElapsedTime(&ms23, ix->ev[2], ix->ev[3]);

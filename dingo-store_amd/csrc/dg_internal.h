@@ -90,6 +90,9 @@ struct dg_index {
   dg_dbuf d_csr_ids;        // [ntotal] i64 (-2 = deleted tombstone)
   dg_dbuf d_csr_vnorms;     // [ntotal] f32 norms in CSR row order
   std::vector<int64_t> h_csr_offsets;  // host copy for planning
+  std::vector<int64_t> h_len_prefix_desc;  // prefix of list lens sorted
+                                           // desc: candidate upper bounds
+  int64_t* h_pinned = nullptr;  // pinned: async alg-bytes readback
   // IVF: column-major 1024-row chunks ([d][nrows_pad] per chunk) — the
   // layout the v2 scan kernel reads
   dg_dbuf d_csr_t;          // transposed chunk data
@@ -177,6 +180,9 @@ void cand_offsets(hipStream_t s, const int32_t* probes, int64_t nq,
 void fill_unit_counts(hipStream_t s, const int32_t* inv_counts, int32_t nlist,
                       const int64_t* csr_offsets, int32_t chunk_rows,
                       int32_t* unit_counts);
+void alg_bytes(hipStream_t s, const int32_t* inv_counts,
+               const int64_t* csr_offsets, int32_t nlist, int64_t row_bytes,
+               int64_t* out /* device scalar, zeroed by caller */);
 void fill_units(hipStream_t s, const int32_t* unit_offsets,
                 const int32_t* inv_counts, int32_t nlist,
                 const int64_t* csr_offsets, int32_t chunk_rows,

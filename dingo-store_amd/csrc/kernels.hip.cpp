@@ -334,6 +334,26 @@ __global__ void k_cand_offsets(const int32_t* __restrict__ probes, int64_t nq,
   q_total[q] = run;
 }
 
+__global__ void k_alg_bytes(const int32_t* __restrict__ inv_counts,
+                            const int64_t* __restrict__ csr_offsets,
+                            int32_t nlist, int64_t row_bytes,
+                            int64_t* __restrict__ out) {
+  __shared__ long long lds[256];
+  long long acc = 0;
+  for (int32_t l = blockIdx.x * blockDim.x + threadIdx.x; l < nlist;
+       l += gridDim.x * blockDim.x)
+    if (inv_counts[l] > 0)
+      acc += (csr_offsets[l + 1] - csr_offsets[l]) * row_bytes;
+  lds[threadIdx.x] = acc;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s; s >>= 1) {
+    if (threadIdx.x < s) lds[threadIdx.x] += lds[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0)
+    atomicAdd((unsigned long long*)out, (unsigned long long)lds[0]);
+}
+
 __global__ void k_unit_counts(const int32_t* __restrict__ inv_counts,
                               int32_t nlist,
                               const int64_t* __restrict__ csr_offsets,
@@ -596,7 +616,6 @@ __device__ __forceinline__ void ivf_scan_col_body(
       static_assert(RPL == 1 || RPL == 2 || RPL % 4 == 0, "RPL");
       float ca[U][RPL], cb[U][RPL];
       const float* colp = col + rr0;
-      const int32_t nfull = d / U;  // d % 4 == 0, U == 4 -> no tail
       auto load_block = [&](float (&c)[U][RPL], int32_t ib) {
 #pragma unroll
         for (int u = 0; u < U; u++) {
@@ -777,6 +796,7 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
   const int64_t row_end = min(list_end, row_start + chunk_rows);
   const int32_t iq0 = inv_offsets[list];
   const int32_t nql = inv_offsets[list + 1] - iq0;
+  if (nql == 0) return;  // unprobed list (all-chunks launch)
   const int wave_id = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int32_t TILE = RPV * WAVE;  // vectors staged per pass
@@ -1455,6 +1475,13 @@ void cand_offsets(hipStream_t s, const int32_t* probes, int64_t nq,
                   int64_t* q_total) {
   hipLaunchKernelGGL(k_cand_offsets, dim3(ceil_div(nq, 128)), dim3(128), 0, s,
                      probes, nq, nprobe, csr_offsets, qp_off, q_total);
+}
+
+void alg_bytes(hipStream_t s, const int32_t* inv_counts,
+               const int64_t* csr_offsets, int32_t nlist, int64_t row_bytes,
+               int64_t* out) {
+  hipLaunchKernelGGL(k_alg_bytes, dim3(16), dim3(256), 0, s, inv_counts,
+                     csr_offsets, nlist, row_bytes, out);
 }
 
 void fill_unit_counts(hipStream_t s, const int32_t* inv_counts, int32_t nlist,

@@ -193,7 +193,7 @@ dg_status dg_sync(dg_index* idx);
 /* ---- range search (SURVEY.md §8f rank 2 — declared now, implemented when
  * row (f2) lands; until then returns DG_ENOT_SUPPORT, which is exactly the
  * reference's brute-force-fallback trigger, vector_reader.cc:1828-1831).
- * CSR out params: caller frees *out_ids/*out_dists with dg_free. ---- */
+ * CSR out params: caller frees *out_ids / *out_dists with dg_free. ---- */
 dg_status dg_range_search(dg_index* idx, int64_t nq, const float* x,
                           float radius, const dg_filter* filter,
                           int64_t* lims /* nq+1 */, int64_t** out_ids,
