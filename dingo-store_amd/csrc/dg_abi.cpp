@@ -1361,7 +1361,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     char* wp = (char*)ix->ws_inv.p;
     int32_t* inv_counts = (int32_t*)wp;            wp += (size_t)nlist * 4;
     int32_t* cursors = (int32_t*)wp;               wp += (size_t)nlist * 4;
-    int32_t* unit_counts = (int32_t*)wp;           wp += (size_t)nlist * 4;
+    wp += (size_t)nlist * 4;  // (reserved; unit counting now launch-free)
     int64_t* inv_offsets64 = (int64_t*)wp;         wp += ((size_t)nlist + 1) * 8;
     int32_t* inv_offsets32 = (int32_t*)wp;         wp += ((size_t)nlist + 1) * 4;
     int32_t* inv_q = (int32_t*)wp;                 wp += (size_t)nq * np * 4;
