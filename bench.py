@@ -21,7 +21,6 @@ oracle's vectorized IVF search on the same structure/host cores — the
 checker timed as a baseline, never the product path.
 """
 import argparse
-import ctypes
 import json
 import os
 import sys

@@ -613,3 +613,29 @@ def test_ivfpq_range_and_cosine():
                               oq, 10, 8)
     frac = ids_match_with_tie_slack(gd2, gi2r, od, oi, tol=5e-3)
     assert frac >= 0.9, frac
+
+
+def test_list_mask_sharding():
+    """dg_set_list_mask (list-sharded deployments): masked search ==
+    oracle with the same mask; the two complementary shards' merge ==
+    whole search."""
+    base, q = make_data(n=20000, d=64, nq=32)
+    nlist, nprobe, k = 32, 16, 10
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        wd, wi = gpu.search(q, k, nprobe=nprobe)
+        mask_a = np.zeros(nlist, np.uint8)
+        mask_a[: nlist // 2] = 1
+        mask_b = 1 - mask_a
+        gpu.set_list_mask(mask_a)
+        ad, ai = gpu.search(q, k, nprobe=nprobe)
+        od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_, q, k, nprobe,
+                                list_mask=mask_a)
+        assert ids_match_with_tie_slack(ad, ai, od, oi) >= 0.98
+        gpu.set_list_mask(mask_b)
+        bd, bi = gpu.search(q, k, nprobe=nprobe)
+        gpu.set_list_mask(None)
+        md, mi = dg.merge_topk(np.stack([ad, bd]), np.stack([ai, bi]), k)
+        assert ids_match_with_tie_slack(md, mi, wd, wi) >= 0.98
+    finally:
+        gpu.close()
