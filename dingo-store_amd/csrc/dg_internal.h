@@ -125,6 +125,11 @@ struct dg_index {
   // workspaces (grown on demand, reused across searches)
   dg_dbuf ws_queries, ws_qnorms, ws_dots, ws_probes, ws_inv, ws_cand,
       ws_units, ws_small, ws_topk;
+  // exclusive-scan block-sum scratch (1024 i64).  Per-index so concurrent
+  // searches of DIFFERENT indexes (multi-region deployment) cannot race on
+  // a process-wide static, and the buffer lives on the index's device
+  // (ADVICE r01 high).  Serialized within an index by search_mu / rw-write.
+  dg_dbuf ws_scan;
 
   // timing
   hipEvent_t ev[12] = {};
@@ -259,8 +264,9 @@ void gather_ids(hipStream_t s, const int64_t* src, const uint32_t* perm,
 void cluster_means(hipStream_t s, const float* grouped, const int64_t* offsets,
                    int32_t nlist, int32_t d, float* centroids);
 void iota_i32(hipStream_t s, int32_t* p, int64_t n, int32_t value);
+// block_sums: caller-owned device scratch of >= 1024 i64 (dg_index::ws_scan)
 void excl_scan_i32_to_i64(hipStream_t s, const int32_t* in, int32_t n,
-                          int64_t* out /* n+1 */);
+                          int64_t* out /* n+1 */, int64_t* block_sums);
 void excl_scan_i64(hipStream_t s, const int64_t* in, int64_t n,
-                   int64_t* out /* n+1 */);
+                   int64_t* out /* n+1 */, int64_t* block_sums);
 }  // namespace dgk

@@ -1295,11 +1295,14 @@ __global__ void k_cluster_means(const float* __restrict__ grouped,
                                 const int64_t* __restrict__ offsets,
                                 int32_t d, float* __restrict__ centroids) {
   // block per cluster; thread t owns dims t, t+256, ...
+  // acc sized for d <= 8192 at blockDim 256 (the create-time ceiling;
+  // acc[8] overflowed for IVF-PQ trains with d > 2048)
   int32_t c = blockIdx.x;
   int64_t s = offsets[c], e = offsets[c + 1];
   if (e <= s) return;  // empty: host handles split
-  double acc[8] = {0};
+  double acc[32] = {0};
   int nd = (d + blockDim.x - 1) / blockDim.x;
+  if (nd > 32) return;  // host enforces d <= 8192
   for (int64_t r = s; r < e; r++) {
     const float* v = grouped + (size_t)r * d;
     for (int i = 0; i < nd; i++) {
@@ -1728,20 +1731,17 @@ void cluster_means(hipStream_t s, const float* grouped, const int64_t* offsets,
 }
 
 void excl_scan_i32_to_i64(hipStream_t s, const int32_t* in, int32_t n,
-                          int64_t* out) {
+                          int64_t* out, int64_t* bs) {
   int nblocks = (int)ceil_div(n + 1, 256);  // thread i==n writes the total
-  static int64_t* bs = nullptr;  // small persistent scratch (max 1023 blocks)
-  if (!bs) (void)hipMalloc(&bs, 1024 * sizeof(int64_t));
   hipLaunchKernelGGL(k_scan_block_i32, dim3(nblocks), dim3(256), 0, s, in, n,
                      out, bs);
   hipLaunchKernelGGL(k_scan_add_offsets, dim3(1), dim3(256), 0, s, out, n, bs,
                      nblocks);
 }
 
-void excl_scan_i64(hipStream_t s, const int64_t* in, int64_t n, int64_t* out) {
+void excl_scan_i64(hipStream_t s, const int64_t* in, int64_t n, int64_t* out,
+                   int64_t* bs) {
   int nblocks = (int)ceil_div(n + 1, 256);
-  static int64_t* bs = nullptr;
-  if (!bs) (void)hipMalloc(&bs, 1024 * sizeof(int64_t));
   hipLaunchKernelGGL(k_scan_block_i64, dim3(nblocks), dim3(256), 0, s, in, n,
                      out, bs);
   hipLaunchKernelGGL(k_scan_add_offsets, dim3(1), dim3(256), 0, s, out,
