@@ -1566,6 +1566,17 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                          cand);
       break;
     }
+    case 9: {  // (12,4) capped to 3 waves/SIMD (uncapped allocates 177)
+      constexpr int QTM = 12;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_col_hi<QTM, 4>),
+                         dim3((uint32_t)n_units), dim3(256), lds, s, units,
+                         csr_offsets, chunk_off, chunk_base, tvec, vnorms,
+                         queries, d, inv_offsets, inv_q, inv_rank, qp_off,
+                         q_cand_base, nprobe, metric, bitmap, chunk_rows,
+                         cand);
+      break;
+    }
     default: DG_SCAN_LAUNCH(16, 4); break;
   }
 #undef DG_SCAN_LAUNCH
