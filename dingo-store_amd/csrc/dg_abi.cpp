@@ -294,7 +294,7 @@ extern "C" void dg_index_destroy(dg_index* ix) {
         &ix->d_codebooks, &ix->d_codes, &ix->d_csr_codes, &ix->d_S,
         &ix->d_cb_norms, &ix->ws_T, &ix->ws_Tf32, &ix->ws_queries, &ix->ws_qnorms,
         &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
-        &ix->ws_small, &ix->ws_topk, &ix->ws_scan})
+        &ix->ws_small, &ix->ws_topk, &ix->ws_scan, &ix->ws_seg})
     dbuf_free(*b);
   for (auto& e : ix->ev)
     if (e) (void)hipEventDestroy(e);
@@ -1010,10 +1010,12 @@ static dg_status finalize_csr(dg_index* ix) {
       return st;
     }
     if (is_ivf &&
-        // +1 chunk of slack: the glds scan's tail quarters issue padded
-        // reads past the last chunk (results discarded at emit)
+        // slack past the last chunk: the glds scan's tail quarters issue
+        // padded reads (d columns) and the asm-pipelined scan runs up to 15
+        // columns ahead of d (last-iteration issue at base+31); results are
+        // never consumed
         (st = dbuf_reserve(ix->d_csr_t,
-                           ((size_t)t_elems + (size_t)d * CR) * 4,
+                           ((size_t)t_elems + (size_t)(d + 16) * CR) * 4,
                            ix->stream, false)) != DG_OK) {
       dbuf_free(rm_tmp);
       return st;
@@ -1245,17 +1247,22 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     int64_t chunk_cols =
         std::max<int64_t>(65536, std::min<int64_t>(N, max_dots / nq));
     int64_t nchunks = (N + chunk_cols - 1) / chunk_cols;
+    // segment the per-chunk select so one block never insert-scans more
+    // than ~8k columns (wide-k path; merged below with select_u64)
+    const int32_t nseg = (int32_t)std::min<int64_t>(
+        16, std::max<int64_t>(1, chunk_cols / 8192));
+    const int64_t slab_k = (int64_t)nchunks * nseg * k;
     if ((st = dbuf_reserve(ix->ws_dots,
                            (size_t)nq * std::min(N, chunk_cols) * 4,
                            ix->stream, false)) != DG_OK ||
         (st = dbuf_reserve(ix->ws_topk,
-                           (size_t)nq * nchunks * k * 8 + (size_t)nq * k * 8 +
+                           (size_t)nq * slab_k * 8 + (size_t)nq * k * 8 +
                                (size_t)(nq + 1) * 16,
                            ix->stream, false)) != DG_OK) {
       return st;
     }
-    uint64_t* slab = (uint64_t*)ix->ws_topk.p;          // nq x nchunks*k
-    uint64_t* final_tk = slab + (size_t)nq * nchunks * k;  // nq x k
+    uint64_t* slab = (uint64_t*)ix->ws_topk.p;          // nq x slab_k
+    uint64_t* final_tk = slab + (size_t)nq * slab_k;    // nq x k
     int64_t* bt = (int64_t*)(final_tk + (size_t)nq * k);   // base/total
     (void)hipEventRecord(ix->ev[1], ix->stream);  // no coarse stage in Flat
     (void)hipEventRecord(ix->ev[2], ix->stream);
@@ -1317,14 +1324,15 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                       d, (float*)ix->ws_dots.p);
       if (st != DG_OK) break;
       dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
-                        (const float*)ix->d_csr_vnorms.p + c0, nq, cc, k,
-                        mode, d_bitmap, c0, slab, nchunks * k, ci * k);
+                        (const float*)ix->d_csr_vnorms.p + c0, nq, cc, cc,
+                        nseg, k, mode, d_bitmap, c0, slab, slab_k,
+                        ci * nseg * k);
     }
     (void)hipEventRecord(ix->ev[3], ix->stream);
     if (st == DG_OK) {
       uint64_t* result = slab;
-      if (nchunks > 1) {
-        dgk::fill_base_total(ix->stream, nq, nchunks * k, bt, bt + nq);
+      if (slab_k > k) {
+        dgk::fill_base_total(ix->stream, nq, slab_k, bt, bt + nq);
         dgk::select_u64(ix->stream, slab, bt, bt + nq, nq, k, final_tk, k);
         result = final_tk;
       }
@@ -1366,9 +1374,11 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
       // exact-ground-truth path recall measurement uses)
       dgk::probes_all(ix->stream, nq, np, maskp, probes);
     } else {
-      if (np > 128) {
-        dg_set_error("nprobe %d > 128 (and < nlist) unsupported this round",
-                     np);
+      if (np > 2048) {
+        // select kernels handle k <= 2048 (LDS-sized per-thread lists);
+        // the reference clamps nprobe only to nlist, so larger nprobe on
+        // huge nlist still fails loudly rather than silently degrading
+        dg_set_error("nprobe %d > 2048 (and < nlist) unsupported", np);
         return DG_ENOT_SUPPORT;
       }
       st = sgemm_dots(ix, dq, nq, (const float*)ix->d_centroids.p, nlist, d,
@@ -1376,10 +1386,34 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
       if (st != DG_OK) {
         return st;
       }
-      dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
-                        (const float*)ix->d_cnorms.p, nq, nlist, np,
-                        metric == DG_METRIC_L2 ? 1 : 2, nullptr, 0, coarse_tk,
-                        np, 0);
+      // segmented coarse top-np: one block per (query, <=8k-col segment),
+      // merged with select_u64 (one block per query); at nlist <= 8192
+      // this is the single-launch path of round 1
+      const int32_t nseg_c = (int32_t)std::min<int64_t>(
+          16, std::max<int64_t>(1, nlist / 8192));
+      if (nseg_c > 1) {
+        dg_dbuf& seg = ix->ws_seg;
+        if ((st = dbuf_reserve(seg,
+                               (size_t)nq * nseg_c * np * 8 +
+                                   (size_t)nq * 2 * 8,
+                               ix->stream, false)) != DG_OK)
+          return st;
+        uint64_t* seg_slab = (uint64_t*)seg.p;
+        int64_t* bt_c = (int64_t*)(seg_slab + (size_t)nq * nseg_c * np);
+        dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
+                          (const float*)ix->d_cnorms.p, nq, nlist, nlist,
+                          nseg_c, np, metric == DG_METRIC_L2 ? 1 : 2,
+                          nullptr, 0, seg_slab, (int64_t)nseg_c * np, 0);
+        dgk::fill_base_total(ix->stream, nq, (int64_t)nseg_c * np, bt_c,
+                             bt_c + nq);
+        dgk::select_u64(ix->stream, seg_slab, bt_c, bt_c + nq, nq, np,
+                        coarse_tk, np);
+      } else {
+        dgk::select_dense(ix->stream, (const float*)ix->ws_dots.p,
+                          (const float*)ix->d_cnorms.p, nq, nlist, nlist, 1,
+                          np, metric == DG_METRIC_L2 ? 1 : 2, nullptr, 0,
+                          coarse_tk, np, 0);
+      }
       dgk::probe_unpack(ix->stream, coarse_tk, nq, np, maskp, probes);
     }
     (void)hipEventRecord(ix->ev[1], ix->stream);
@@ -1549,8 +1583,10 @@ extern "C" dg_status dg_search_device(dg_index* ix, int64_t nq,
     return DG_EINVAL;
   }
   if (k <= 0) return DG_OK;  // reference: topk <= 0 => OK no-op
-  if (k > 128) {
-    dg_set_error("k > 128 not supported this round");
+  if (k > 2048) {
+    // per-thread LDS top-k lists cap out at k=2048 (select_threads);
+    // far above the reference's vector_max_batch_count-scale top_n
+    dg_set_error("k > 2048 not supported");
     return DG_ENOT_SUPPORT;
   }
   DeviceGuard g(ix->device);

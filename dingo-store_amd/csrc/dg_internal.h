@@ -130,6 +130,8 @@ struct dg_index {
   // a process-wide static, and the buffer lives on the index's device
   // (ADVICE r01 high).  Serialized within an index by search_mu / rw-write.
   dg_dbuf ws_scan;
+  // segmented-select slab (wide-k coarse top-nprobe merge)
+  dg_dbuf ws_seg;
 
   // timing
   hipEvent_t ev[12] = {};
@@ -152,10 +154,15 @@ void probes_all(hipStream_t s, int64_t nq, int32_t nprobe,
                 const uint8_t* mask, int32_t* probes);
 void init_cursors(hipStream_t s, const int64_t* offsets, int32_t n,
                   int32_t* cursors);
+// top-k per row over a dense scores matrix.  nseg > 1 splits columns into
+// nseg segments (one block each, k results per segment at out_offset+seg*k;
+// caller merges with select_u64).  ld = row stride of scores.
 void select_dense(hipStream_t s, const float* scores, const float* cnorms,
-                  int64_t rows, int64_t cols, int32_t k, int mode,
-                  const uint32_t* bitmap, int64_t col_base, uint64_t* out,
-                  int64_t out_stride, int64_t out_offset);
+                  int64_t rows, int64_t cols, int64_t ld, int32_t nseg,
+                  int32_t k, int mode, const uint32_t* bitmap,
+                  int64_t col_base, uint64_t* out, int64_t out_stride,
+                  int64_t out_offset);
+int select_dense_threads(int32_t k);
 void select_u64(hipStream_t s, const uint64_t* cand, const int64_t* base,
                 const int64_t* total, int64_t nq, int32_t k, uint64_t* out,
                 int64_t out_stride);

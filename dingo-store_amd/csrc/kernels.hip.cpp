@@ -202,30 +202,41 @@ __device__ void topk_merge_block(uint64_t* lds, int32_t k, int32_t cnt,
 }
 
 // dense scores row scan: mode 0 key=score, 1 key=cnorm[col]-2*score,
-// 2 key=-score.  col_base added to the packed payload.
+// 2 key=-score.  col_base added to the packed payload (and is the global
+// column index for the bitmap); cnorms is indexed by the local column.
+// ld = row stride of scores.  blockIdx.y picks a seg_w-wide column segment
+// (the wide-k path runs (rows x S) blocks, writing k per segment at
+// out_offset + y*k, merged afterwards with select_u64); plain callers
+// launch gridDim.y = 1 with seg_w = cols.
 __global__ void k_select_dense(const float* __restrict__ scores,
                                const float* __restrict__ cnorms, int64_t rows,
-                               int64_t cols, int32_t k, int mode,
+                               int64_t cols, int64_t ld, int64_t seg_w,
+                               int32_t k, int mode,
                                const uint32_t* __restrict__ bitmap,
                                int64_t col_base,
-                               uint64_t* __restrict__ out /* rows x k */,
+                               uint64_t* __restrict__ out,
                                int64_t out_stride, int64_t out_offset) {
   extern __shared__ uint64_t lds[];
   int64_t row = blockIdx.x;
   if (row >= rows) return;
-  const float* sr = scores + row * cols;
+  const int64_t seg_start = (int64_t)blockIdx.y * seg_w;
+  const int64_t cseg = min(seg_w, cols - seg_start);
+  const float* sr = scores + row * ld + seg_start;
+  const float* cn = cnorms + seg_start;
   uint64_t* mine = lds + (size_t)threadIdx.x * k;
   int32_t cnt = 0;
-  for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+  for (int64_t c = threadIdx.x; c < cseg; c += blockDim.x) {
     if (bitmap) {
-      int64_t g = col_base + c;
+      int64_t g = col_base + seg_start + c;
       if (!((bitmap[g >> 5] >> (g & 31)) & 1)) continue;
     }
     float s = sr[c];
-    float key = (mode == 0) ? s : (mode == 1) ? cnorms[c] - 2.0f * s : -s;
-    topk_insert(mine, cnt, k, pack_cand(key, (uint32_t)(col_base + c)));
+    float key = (mode == 0) ? s : (mode == 1) ? cn[c] - 2.0f * s : -s;
+    topk_insert(mine, cnt, k,
+                pack_cand(key, (uint32_t)(col_base + seg_start + c)));
   }
-  topk_merge_block(lds, k, cnt, out + row * out_stride + out_offset);
+  topk_merge_block(lds, k, cnt,
+                   out + row * out_stride + out_offset + blockIdx.y * k);
 }
 
 // candidate (u64) segment scan per query
@@ -777,8 +788,8 @@ __global__ void k_f32_to_f16(const float* __restrict__ in, int64_t n,
 // vector and walks m = 0..M-1 sequentially, so all lanes of a wave gather
 // within the same 1 KB rows of T[q] and S[l] (L1-resident after first
 // touch).  No cross-lane reduction.
-template <int RPV>  // rows (vectors) per lane; TILE = RPV * 64
-__global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
+template <int RPV, int MINB = 1>  // rows (vectors) per lane; TILE = RPV * 64
+__global__ void __launch_bounds__(256, MINB) k_ivfpq_scan(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const uint8_t* __restrict__ csr_codes, const __half* __restrict__ S,
     const __half* __restrict__ T, const float* __restrict__ coarse_dots,
@@ -894,6 +905,187 @@ __global__ void __launch_bounds__(256, 6) k_ivfpq_scan(
           if (bitmap) pass = (bitmap[rw >> 5] >> (rw & 31)) & 1;
           cand[cb0 + rw] =
               pass ? pack_cand(acc[r], (uint32_t)rw) : kCandEmpty;
+        }
+      }
+    }
+  }
+}
+
+// ---------- asm-pipelined columnar scan (v3, DG_SCAN_VARIANT=10) ----------
+// Round-2 redesign of the dim loop.  The C++ double-buffer rotation of v2
+// compiles to a register-allocator artifact: hipcc copies each loaded
+// float4 into the loop-carried phi registers THROUGH an s_waitcnt vmcnt(0)
+// placed right after the first load of the next block — the "pipeline"
+// waits for the load it just issued, exposing full HBM latency (~900 cy)
+// every 8 dims.  Here the loads are inline-asm global_load_dwordx4 into
+// explicitly tied buffers with counted s_waitcnt vmcnt(N) (the compiler
+// does not know the loads are pending, so it cannot insert its own waits),
+// modulo-scheduled over DEPTH=4 stages x U=4 dims: 16 column loads stay in
+// flight per lane while one stage computes — outstanding bytes per CU
+// (8 waves x ~12 KB) far exceed the ~7 KB Little's-law need at the 6.3 TB/s
+// achievable ceiling.  vmcnt discipline: the dim loop contains NO other
+// vmem (a vmcnt(0) drain precedes the prologue so emit-stores of the
+// previous tile don't skew the counts); loads past d land in d_csr_t's
+// one-chunk slack and are never consumed.
+typedef float dg_f4 __attribute__((ext_vector_type(4)));
+
+#define DG_GLOAD4(dst, ptr)                                         \
+  asm volatile("global_load_dwordx4 %0, %1, off"                    \
+               : "=v"(dst)                                          \
+               : "v"((const void*)(ptr)))
+// counted wait, tying the stage's 4 buffers so their consumers cannot be
+// scheduled above the wait
+#define DG_WAITV(n, b)                                              \
+  asm volatile("s_waitcnt vmcnt(" #n ")"                            \
+               : "+v"((b)[0]), "+v"((b)[1]), "+v"((b)[2]), "+v"((b)[3]))
+
+template <int QTM>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  constexpr int RPL = 4;  // rows per lane (float4 column loads)
+  constexpr int U = 4;    // dims per stage
+  extern __shared__ __attribute__((aligned(16))) float smem[];  // [QTM * d]
+  int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);          // [QTM]
+
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* col = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+  const int64_t row0 = list_start + (int64_t)chunk * chunk_rows;
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  __builtin_assume(d % 4 == 0 && d > 0);
+
+  for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
+    const int32_t qt = min(QTM, nql - t0);
+    __syncthreads();
+    for (int32_t j = 0; j < qt; j++) {
+      int32_t q = inv_q[iq0 + t0 + j];
+      const float4* src = (const float4*)(queries + (size_t)q * d);
+      float4* dst = (float4*)(smem + (size_t)j * d);
+      for (int i = threadIdx.x; i < d / 4; i += blockDim.x) dst[i] = src[i];
+    }
+    for (size_t i = (size_t)qt * d + threadIdx.x; i < (size_t)QTM * d;
+         i += blockDim.x)
+      smem[i] = 0.f;
+    if (threadIdx.x < QTM) {
+      int32_t j = threadIdx.x;
+      if (j < qt) {
+        int32_t q = inv_q[iq0 + t0 + j];
+        int32_t rank = inv_rank[iq0 + t0 + j];
+        cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                   list_start;
+      } else {
+        cbase[j] = 0;
+      }
+    }
+    __syncthreads();
+
+    for (int32_t rb = wave_id * WAVE * RPL; rb < nrows_pad;
+         rb += 4 * WAVE * RPL) {
+      const int32_t rr0 = rb + lane * RPL;
+      if (rr0 >= nrows_pad) continue;
+      float acc[QTM][RPL];
+#pragma unroll
+      for (int j = 0; j < QTM; j++)
+#pragma unroll
+        for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
+
+      // drain prior vmem (previous tile's emit stores) so vmcnt counts
+      // below track ONLY this loop's column loads
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+      const char* colp = (const char*)(col + rr0);
+      const size_t cstride = (size_t)nrows_pad * 4;  // bytes per dim column
+      dg_f4 b0[U], b1[U], b2[U], b3[U];
+      auto issue = [&](dg_f4 (&b)[U], int32_t ib) {
+        const char* p = colp + (size_t)ib * cstride;
+#pragma unroll
+        for (int u = 0; u < U; u++) DG_GLOAD4(b[u], p + u * cstride);
+      };
+      auto compute = [&](dg_f4 (&b)[U], int32_t ib) {
+#pragma unroll
+        for (int j = 0; j < QTM; j++) {
+          const float* qp = smem + (size_t)j * d + ib;
+          const float2 qa = *(const float2*)qp;
+          const float2 qb = *(const float2*)(qp + 2);
+          const float qv[4] = {qa.x, qa.y, qb.x, qb.y};
+#pragma unroll
+          for (int u = 0; u < U; u++)
+#pragma unroll
+            for (int x = 0; x < RPL; x++) acc[j][x] += b[u][x] * qv[u];
+        }
+      };
+      // modulo schedule, period 4 stages = 16 dims
+      issue(b0, 0);
+      issue(b1, U);
+      issue(b2, 2 * U);
+      issue(b3, 3 * U);
+      int32_t base = 0;
+      for (; base + 16 < d; base += 16) {
+        DG_WAITV(12, b0);
+        compute(b0, base);
+        issue(b0, base + 16);
+        DG_WAITV(12, b1);
+        compute(b1, base + 4);
+        issue(b1, base + 20);
+        DG_WAITV(12, b2);
+        compute(b2, base + 8);
+        issue(b2, base + 24);
+        DG_WAITV(12, b3);
+        compute(b3, base + 12);
+        issue(b3, base + 28);
+      }
+      {  // epilogue: r = d - base in {4, 8, 12, 16} (d % 4 == 0)
+        const int32_t r = d - base;
+        DG_WAITV(12, b0);
+        compute(b0, base);
+        if (r > 4) {
+          DG_WAITV(8, b1);
+          compute(b1, base + 4);
+        }
+        if (r > 8) {
+          DG_WAITV(4, b2);
+          compute(b2, base + 8);
+        }
+        if (r > 12) {
+          DG_WAITV(0, b3);
+          compute(b3, base + 12);
+        }
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+#pragma unroll
+      for (int j = 0; j < QTM; j++) {
+        if (j < qt) {
+          const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+          for (int x = 0; x < RPL; x++) {
+            const int32_t rl = rr0 + x;
+            if (rl < nrows) {
+              const int64_t r = row0 + rl;
+              bool pass = true;
+              if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+              float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                        : -acc[j][x];
+              cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+            }
+          }
         }
       }
     }
@@ -1411,22 +1603,28 @@ void build_pass_bitmap(hipStream_t s, const int64_t* ids, int64_t n,
                      dim3(256), 0, s, ids, n, *f, bitmap);
 }
 
-// choose block size by k: T*k*8*2 <= 128KB
+// choose block size by k: 2*T*k*8 <= 128 KB dynamic LDS (measured-good on
+// gfx950).  Blocks shrink below a wave for the rare huge-k calls — correct,
+// slow, and documented; host validates k <= 2048.
 static inline int select_threads(int32_t k) {
-  if (k <= 32) return 256;
-  if (k <= 128) return 64;
-  return 0;  // host validates k <= 128
+  int T = (k <= 32) ? 256 : 64;
+  while (T > 2 && 2ull * T * k * 8 > (128u << 10)) T >>= 1;
+  return T;
 }
 
+int select_dense_threads(int32_t k) { return select_threads(k); }
+
 void select_dense(hipStream_t s, const float* scores, const float* cnorms,
-                  int64_t rows, int64_t cols, int32_t k, int mode,
-                  const uint32_t* bitmap, int64_t col_base, uint64_t* out,
-                  int64_t out_stride, int64_t out_offset) {
+                  int64_t rows, int64_t cols, int64_t ld, int32_t nseg,
+                  int32_t k, int mode, const uint32_t* bitmap,
+                  int64_t col_base, uint64_t* out, int64_t out_stride,
+                  int64_t out_offset) {
   int T = select_threads(k);
   size_t lds = 2ull * T * k * 8;
-  hipLaunchKernelGGL(k_select_dense, dim3((uint32_t)rows), dim3(T), lds, s,
-                     scores, cnorms, rows, cols, k, mode, bitmap, col_base,
-                     out, out_stride, out_offset);
+  int64_t seg_w = (cols + nseg - 1) / nseg;
+  hipLaunchKernelGGL(k_select_dense, dim3((uint32_t)rows, (uint32_t)nseg),
+                     dim3(T), lds, s, scores, cnorms, rows, cols, ld, seg_w,
+                     k, mode, bitmap, col_base, out, out_stride, out_offset);
 }
 
 void select_u64(hipStream_t s, const uint64_t* cand, const int64_t* base,
@@ -1566,6 +1764,26 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                          cand);
       break;
     }
+    case 10: {  // asm-pipelined counted-vmcnt scan (v3)
+      constexpr int QTM = 16;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_pipe<QTM>), dim3((uint32_t)n_units),
+                         dim3(256), lds, s, units, csr_offsets, chunk_off,
+                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
+                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
+                         metric, bitmap, chunk_rows, cand);
+      break;
+    }
+    case 11: {  // asm-pipelined, QTM=12
+      constexpr int QTM = 12;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_pipe<QTM>), dim3((uint32_t)n_units),
+                         dim3(256), lds, s, units, csr_offsets, chunk_off,
+                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
+                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
+                         metric, bitmap, chunk_rows, cand);
+      break;
+    }
     case 9: {  // (12,4) capped to 3 waves/SIMD (uncapped allocates 177)
       constexpr int QTM = 12;
       size_t lds = (size_t)QTM * d * 4 + QTM * 8;
@@ -1664,25 +1882,37 @@ void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
                 int32_t nprobe, int metric, const uint32_t* bitmap,
                 int32_t chunk_rows, uint64_t* cand) {
   if (!n_units) return;
-  static int rpv = []() {
-    const char* e = getenv("DG_PQ_RPV");
-    return e ? atoi(e) : 4;
-  }();
-  if (rpv == 8) {  // TILE=512: halves per-tile T-row re-fetches, 1 block/CU
-    size_t lds = 512 * (size_t)M + (size_t)M * 256 * 2;
-    hipLaunchKernelGGL((k_ivfpq_scan<8>), dim3((uint32_t)n_units), dim3(256),
-                       lds, s, units, csr_offsets, csr_codes, S, T,
-                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
-                       qp_off, q_cand_base, nprobe, metric, bitmap,
-                       chunk_rows, cand);
-  } else {
-    size_t lds = 256 * (size_t)M + (size_t)M * 256 * 2;  // codes + f16 S_l
-    hipLaunchKernelGGL((k_ivfpq_scan<4>), dim3((uint32_t)n_units), dim3(256),
-                       lds, s, units, csr_offsets, csr_codes, S, T,
-                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
-                       qp_off, q_cand_base, nprobe, metric, bitmap,
-                       chunk_rows, cand);
+  // T-table (L2/L3) traffic per chunk scales as 1/TILE: every staged tile
+  // walks each probing query's 48 KB T window once, so the largest tile
+  // whose codes + S_l fit LDS wins unless occupancy dominates (A/B'd;
+  // DG_PQ_RPV forces a tile for experiments).  TILE = RPV * 64 rows.
+  const char* e = getenv("DG_PQ_RPV");
+  int rpv = e ? atoi(e) : 0;
+  if (rpv != 4 && rpv != 8 && rpv != 16) {
+    rpv = 16;
+    while (rpv > 4 &&
+           (size_t)rpv * 64 * M + (size_t)M * 256 * 2 > (158u << 10))
+      rpv >>= 1;
   }
+  size_t lds = (size_t)rpv * 64 * M + (size_t)M * 256 * 2;  // codes + f16 S_l
+  if (rpv == 16)
+    hipLaunchKernelGGL((k_ivfpq_scan<16, 1>), dim3((uint32_t)n_units), dim3(256),
+                       lds, s, units, csr_offsets, csr_codes, S, T,
+                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
+                       qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
+  else if (rpv == 8)
+    hipLaunchKernelGGL((k_ivfpq_scan<8, 1>), dim3((uint32_t)n_units), dim3(256),
+                       lds, s, units, csr_offsets, csr_codes, S, T,
+                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
+                       qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
+  else
+    hipLaunchKernelGGL((k_ivfpq_scan<4, 6>), dim3((uint32_t)n_units), dim3(256),
+                       lds, s, units, csr_offsets, csr_codes, S, T,
+                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
+                       qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
 }
 
 void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
