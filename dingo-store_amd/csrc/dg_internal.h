@@ -40,6 +40,15 @@
 
 void dg_set_error(const char* fmt, ...);
 
+// Internal ingestion used by the faiss-file loader (faiss_io.cpp): uploads
+// rows with their FILE-stored coarse assignment (recomputing on GPU could
+// flip fp ties vs the CPU faiss reading the same file) and without
+// re-normalizing (file data is already in stored form).  x for Flat/IVF,
+// codes for PQ.  Takes the write lock itself.
+dg_status dg_ingest_rows(dg_index* ix, int64_t n, const int64_t* ids,
+                         const float* x, const uint8_t* codes,
+                         const int32_t* assign);
+
 // Device-side filter descriptor (POD copied into kernel args).
 struct dg_dev_filter {
   int32_t kind;    // dg_filter_kind

@@ -1,3 +1,5 @@
+set -x
+echo AB3_START
 cd /root/repo
 echo "=== parity (ivf subset) variant 10 ==="
 DG_SCAN_VARIANT=10 timeout 600 python -m pytest tests/test_gpu_parity.py -x -q -m gpu -k "ivf and not pq" 2>&1 | tail -3
