@@ -813,6 +813,54 @@ extern "C" dg_status dg_add(dg_index* ix, int64_t n, const int64_t* ids,
   return add_impl(ix, n, ids, x, false);
 }
 
+// faiss-file loader ingestion (dg_internal.h): rows arrive with their
+// file-stored coarse assignment and in stored (already-normalized) form.
+dg_status dg_ingest_rows(dg_index* ix, int64_t n, const int64_t* ids,
+                         const float* x, const uint8_t* codes,
+                         const int32_t* assign) {
+  if (!ix || !ids || n <= 0 || (!x && !codes)) return DG_EINVAL;
+  std::unique_lock lk(ix->rw);
+  DeviceGuard g(ix->device);
+  const int32_t d = ix->desc.d;
+  const int64_t n0 = ix->ntotal;
+  const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
+  dg_status st = DG_OK;
+  if (!is_pq)
+    st = dbuf_reserve(ix->d_vectors, (size_t)(n0 + n) * d * 4, ix->stream,
+                      true);
+  if (st == DG_OK)
+    st = dbuf_reserve(ix->d_ids, (size_t)(n0 + n) * 8, ix->stream, true);
+  if (st == DG_OK)
+    st = dbuf_reserve(ix->d_assign, (size_t)(n0 + n) * 4, ix->stream, true);
+  if (st == DG_OK && is_pq)
+    st = dbuf_reserve(ix->d_codes, (size_t)(n0 + n) * ix->desc.pq_m,
+                      ix->stream, true);
+  if (st != DG_OK) return st;
+  if (!is_pq)
+    DG_HIP_CHECK(hipMemcpyAsync((float*)ix->d_vectors.p + (size_t)n0 * d, x,
+                                (size_t)n * d * 4, hipMemcpyHostToDevice,
+                                ix->stream));
+  else
+    DG_HIP_CHECK(hipMemcpyAsync(
+        (uint8_t*)ix->d_codes.p + (size_t)n0 * ix->desc.pq_m, codes,
+        (size_t)n * ix->desc.pq_m, hipMemcpyHostToDevice, ix->stream));
+  DG_HIP_CHECK(hipMemcpyAsync((int64_t*)ix->d_ids.p + n0, ids, (size_t)n * 8,
+                              hipMemcpyHostToDevice, ix->stream));
+  if (assign) {
+    DG_HIP_CHECK(hipMemcpyAsync((int32_t*)ix->d_assign.p + n0, assign,
+                                (size_t)n * 4, hipMemcpyHostToDevice,
+                                ix->stream));
+  } else {
+    dgk::iota_i32(ix->stream, (int32_t*)ix->d_assign.p + n0, n, 0);
+  }
+  DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
+  for (int64_t i = 0; i < n; i++)
+    if (ids[i] >= 0) ix->id_count.emplace(ids[i], 1);
+  ix->ntotal += n;
+  ix->csr_valid = false;
+  return DG_OK;
+}
+
 extern "C" dg_status dg_add_device(dg_index* ix, int64_t n,
                                    const int64_t* ids, const float* d_x) {
   return add_impl(ix, n, ids, d_x, true);

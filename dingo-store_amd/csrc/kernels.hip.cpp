@@ -929,10 +929,13 @@ __global__ void __launch_bounds__(256, MINB) k_ivfpq_scan(
 // one-chunk slack and are never consumed.
 typedef float dg_f4 __attribute__((ext_vector_type(4)));
 
+// NOTE: the address must be bound as uint64_t — a pointer-typed "v"
+// operand binds a single 32-bit VGPR and truncates the VA (measured: GPU
+// memory fault at the truncated low-32-bit address)
 #define DG_GLOAD4(dst, ptr)                                         \
   asm volatile("global_load_dwordx4 %0, %1, off"                    \
                : "=v"(dst)                                          \
-               : "v"((const void*)(ptr)))
+               : "v"((uint64_t)(uintptr_t)(ptr)))
 // counted wait, tying the stage's 4 buffers so their consumers cannot be
 // scheduled above the wait
 #define DG_WAITV(n, b)                                              \

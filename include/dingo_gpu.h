@@ -190,9 +190,10 @@ dg_status dg_search_device(dg_index* idx, int64_t nq, const float* d_x,
                            float* d_out_dist, int64_t* d_out_ids);
 dg_status dg_sync(dg_index* idx);
 
-/* ---- range search (SURVEY.md §8f rank 2 — declared now, implemented when
- * row (f2) lands; until then returns DG_ENOT_SUPPORT, which is exactly the
- * reference's brute-force-fallback trigger, vector_reader.cc:1828-1831).
+/* ---- range search ----
+ * Restates VectorIndexIvfFlat::RangeSearch (vector_index_ivf_flat.cc:278-368):
+ * radius in faiss convention (L2: dist < radius, IP/cos: score > radius; the
+ * C++ shim applies the 1-r flip of :302-305); results best-first per query.
  * CSR out params: caller frees *out_ids / *out_dists with dg_free. ---- */
 dg_status dg_range_search(dg_index* idx, int64_t nq, const float* x,
                           float radius, const dg_filter* filter,
@@ -202,9 +203,30 @@ void dg_free(void* p);
 
 /* ---- persistence (Save/Load, vector_index.h:168-170; snapshot files,
  * src/vector/vector_index_snapshot_manager.cc:583-599).  Own container
- * format v1 (documented in DESIGN.md); faiss-file converter is §8f rank 1. */
+ * format v1 (documented in DESIGN.md) plus the faiss-compatible container
+ * below. */
 dg_status dg_save(dg_index* idx, const char* path);
 dg_status dg_load(dg_index** out, const char* path, int32_t device);
+
+/* ---- faiss-file-compatible persistence ----
+ * The reference's snapshot/install cycle ships faiss::write_index files
+ * (vector_index_snapshot_manager.cc:583-599; written at
+ * vector_index_flat.cc:354, vector_index_ivf_flat.cc:398, loaded at
+ * vector_index_flat.cc:368-462, vector_index_ivf_flat.cc:413-514).  These
+ * functions emit/ingest the faiss 1.7.x container byte layout (the
+ * reference's faiss fork is >= 1.7.3 — it uses faiss::SearchParameters,
+ * vector_index_flat.cc:232):
+ *   FLAT     -> IndexIDMap2{IndexFlatL2|IndexFlatIP}   ("IxM2"/"IxF2"/"IxFI")
+ *   IVF_FLAT -> IndexIVFFlat                            ("IwFl")
+ *   IVF_PQ   -> IndexIVFPQ (residual, nbits=8)          ("IwPQ")
+ * Cosine indexes are written as IP over stored (normalized) vectors, like
+ * the reference (vector_index_flat.cc:88-91); on load, pass
+ * metric_override = DG_METRIC_COSINE to reconstruct a cosine index from an
+ * IP-metric file (-1 keeps the file metric).  Tombstoned rows are dropped
+ * at save.  Format notes and the byte-level layout are in DESIGN.md. */
+dg_status dg_save_faiss(dg_index* idx, const char* path);
+dg_status dg_load_faiss(dg_index** out, const char* path,
+                        int32_t metric_override, int32_t device);
 
 /* ---- multi-GPU sharding support ----
  * Each rank holds a shard of the database (row-sharded)
