@@ -1936,6 +1936,19 @@ extern "C" dg_status dg_load(dg_index** out, const char* path,
   return DG_OK;
 }
 
+// ---------------- wrapper-lifecycle lock ----------------
+// LockWrite/UnlockWrite (vector_index.h:192-193): the exclusive lock the
+// reference wrapper takes around its fork-save window; must be released by
+// the locking thread (std::shared_mutex requirement, matching bthread
+// RWLock usage in the reference).
+extern "C" void dg_lock_write(dg_index* ix) {
+  if (ix) ix->rw.lock();
+}
+
+extern "C" void dg_unlock_write(dg_index* ix) {
+  if (ix) ix->rw.unlock();
+}
+
 // ---------------- stats ----------------
 extern "C" dg_status dg_stats(dg_index* ix, dg_stats_out* out) {
   if (!ix || !out) return DG_EINVAL;
@@ -1956,6 +1969,7 @@ extern "C" dg_status dg_stats(dg_index* ix, dg_stats_out* out) {
                  &ix->d_cb_norms})
     db += b->cap;
   out->device_bytes = (int64_t)db;
+  out->deleted_count = ix->n_deleted;
   if (ix->times.last_nq > 0) {
     // hold search_mu so a concurrent dg_search cannot re-record the events
     // or overwrite the pinned alg-bytes scalar mid-read

@@ -931,18 +931,30 @@ typedef float dg_f4 __attribute__((ext_vector_type(4)));
 
 // NOTE: the address must be bound as uint64_t — a pointer-typed "v"
 // operand binds a single 32-bit VGPR and truncates the VA (measured: GPU
-// memory fault at the truncated low-32-bit address)
+// memory fault at the truncated low-32-bit address).  "=&v" (early
+// clobber) keeps the allocator from overlapping the destination tuple
+// with the address pair: the hardware does not interlock VALU writes to
+// an ASYNC load's destination, and the compiler cannot see that the asm
+// load is still in flight.
 #define DG_GLOAD4(dst, ptr)                                         \
   asm volatile("global_load_dwordx4 %0, %1, off"                    \
-               : "=v"(dst)                                          \
+               : "=&v"(dst)                                         \
                : "v"((uint64_t)(uintptr_t)(ptr)))
 // counted wait, tying the stage's 4 buffers so their consumers cannot be
-// scheduled above the wait
+// scheduled above the wait.  SAFE mode drains everything (diagnostic
+// baseline for the counted schedule).
 #define DG_WAITV(n, b)                                              \
   asm volatile("s_waitcnt vmcnt(" #n ")"                            \
                : "+v"((b)[0]), "+v"((b)[1]), "+v"((b)[2]), "+v"((b)[3]))
+#define DG_WAITV_N(n, b)                                            \
+  do {                                                              \
+    if (SAFE)                                                       \
+      DG_WAITV(0, b);                                               \
+    else                                                            \
+      DG_WAITV(n, b);                                               \
+  } while (0)
 
-template <int QTM>
+template <int QTM, bool SAFE = false>
 __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
@@ -1041,29 +1053,29 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
       issue(b3, 3 * U);
       int32_t base = 0;
       for (; base + 16 < d; base += 16) {
-        DG_WAITV(12, b0);
+        DG_WAITV_N(12, b0);
         compute(b0, base);
         issue(b0, base + 16);
-        DG_WAITV(12, b1);
+        DG_WAITV_N(12, b1);
         compute(b1, base + 4);
         issue(b1, base + 20);
-        DG_WAITV(12, b2);
+        DG_WAITV_N(12, b2);
         compute(b2, base + 8);
         issue(b2, base + 24);
-        DG_WAITV(12, b3);
+        DG_WAITV_N(12, b3);
         compute(b3, base + 12);
         issue(b3, base + 28);
       }
       {  // epilogue: r = d - base in {4, 8, 12, 16} (d % 4 == 0)
         const int32_t r = d - base;
-        DG_WAITV(12, b0);
+        DG_WAITV_N(12, b0);
         compute(b0, base);
         if (r > 4) {
-          DG_WAITV(8, b1);
+          DG_WAITV_N(8, b1);
           compute(b1, base + 4);
         }
         if (r > 8) {
-          DG_WAITV(4, b2);
+          DG_WAITV_N(4, b2);
           compute(b2, base + 8);
         }
         if (r > 12) {
@@ -1775,6 +1787,17 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                          chunk_base, tvec, vnorms, queries, d, inv_offsets,
                          inv_q, inv_rank, qp_off, q_cand_base, nprobe,
                          metric, bitmap, chunk_rows, cand);
+      break;
+    }
+    case 12: {  // asm-pipelined, full-drain waits (correctness baseline)
+      constexpr int QTM = 16;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_pipe<QTM, true>),
+                         dim3((uint32_t)n_units), dim3(256), lds, s, units,
+                         csr_offsets, chunk_off, chunk_base, tvec, vnorms,
+                         queries, d, inv_offsets, inv_q, inv_rank, qp_off,
+                         q_cand_base, nprobe, metric, bitmap, chunk_rows,
+                         cand);
       break;
     }
     case 11: {  // asm-pipelined, QTM=12

@@ -52,6 +52,7 @@ class _Stats(C.Structure):
         ("last_nq", C.c_int64),
         ("last_scan_bytes_algorithmic", C.c_int64),
         ("last_scan_gbps_algorithmic", C.c_double),
+        ("deleted_count", C.c_int64),
     ]
 
 
@@ -295,11 +296,36 @@ class Index:
     def load(cls, path, device=-1):
         h = C.c_void_p()
         _check(lib().dg_load(C.byref(h), path.encode(), device), "dg_load")
+        return cls._from_handle(h)
+
+    def save_faiss(self, path):
+        """Write a faiss-1.7.x-compatible container (the snapshot format the
+        reference ships between nodes, vector_index_snapshot_manager.cc)."""
+        l = lib()
+        l.dg_save_faiss.argtypes = [C.c_void_p, C.c_char_p]
+        _check(l.dg_save_faiss(self.h, path.encode()), "dg_save_faiss")
+
+    @classmethod
+    def load_faiss(cls, path, metric=-1, device=-1):
+        """Load a faiss container (IxM2{IndexFlat} / IwFl / IwPQ).  metric
+        = COSINE reinterprets an IP-metric file as a cosine index (the
+        reference stores cosine as IP over normalized vectors)."""
+        l = lib()
+        l.dg_load_faiss.argtypes = [C.POINTER(C.c_void_p), C.c_char_p,
+                                    C.c_int32, C.c_int32]
+        h = C.c_void_p()
+        _check(l.dg_load_faiss(C.byref(h), path.encode(), metric, device),
+               "dg_load_faiss")
+        return cls._from_handle(h)
+
+    @classmethod
+    def _from_handle(cls, h):
         idx = cls.__new__(cls)
         idx.h = h
         st = idx.stats()
         idx.kind, idx.metric = st["kind"], st["metric"]
         idx.d, idx.nlist = st["d"], st["nlist"]
+        idx.m = 0
         return idx
 
     def stats(self):

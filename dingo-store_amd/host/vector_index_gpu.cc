@@ -9,7 +9,22 @@
 #include <cmath>
 #include <cstring>
 
+#include <algorithm>
+#include <unordered_set>
+
 namespace dingogpu {
+
+ConcreteFilterFunctor::ConcreteFilterFunctor(const std::vector<int64_t>& ids,
+                                             bool is_negation)
+    : sorted_(ids), negation_(is_negation) {
+  std::sort(sorted_.begin(), sorted_.end());
+  sorted_.erase(std::unique(sorted_.begin(), sorted_.end()), sorted_.end());
+}
+
+bool ConcreteFilterFunctor::Check(int64_t id) {
+  bool found = std::binary_search(sorted_.begin(), sorted_.end(), id);
+  return negation_ ? !found : found;
+}
 
 bool SortFilterFunctor::Check(int64_t id) {
   int64_t lo = 0, hi = (int64_t)ids_.size() - 1;
@@ -121,18 +136,61 @@ class GpuIndexBase : public VectorIndex {
     std::vector<int64_t> ids;
     Status s = pack(v, dim_, x, ids);
     if (!s.ok()) return s;
-    return from_dg(dg_train(idx_, (int64_t)v.size(), x.data()));
+    s = from_dg(dg_train(idx_, (int64_t)v.size(), x.data()));
+    if (s.ok()) train_data_size_ = (int64_t)v.size();
+    return s;
+  }
+  Status Train(std::vector<float>& train_datas) override {
+    // raw-float Train (vector_index.h:196; size checks per
+    // vector_index_ivf_flat.cc:644-652)
+    size_t data_size = train_datas.size() / dim_;
+    if (data_size == 0)
+      return {kEillegalParamteters, "data size invalid"};
+    if (train_datas.size() % dim_ != 0)
+      return {kEillegalParamteters, "dimension not match"};
+    Status s = from_dg(dg_train(idx_, (int64_t)data_size,
+                                train_datas.data()));
+    if (s.ok()) train_data_size_ = (int64_t)data_size;
+    return s;
   }
   bool IsTrained() override {
     dg_stats_out s{};
     return dg_stats(idx_, &s) == DG_OK && s.is_trained;
   }
+  Status GetDeletedCount(int64_t& deleted_count) override {
+    // the reference's faiss path compacts on remove and reports 0
+    // (vector_index_ivf_flat.cc:539-542); here removes tombstone until the
+    // next finalize, so the pending count is reported
+    dg_stats_out s{};
+    dg_status st = dg_stats(idx_, &s);
+    deleted_count = s.deleted_count;
+    return from_dg(st);
+  }
+  bool IsExceedsMaxElements(int64_t) override {
+    return false;  // flat.cc:510-512, ivf_flat.cc:573-575
+  }
+  bool NeedToRebuild() override { return false; }  // per-kind overrides
+  bool NeedToSave(int64_t last_save_log_behind) override {
+    // flat.cc:515-531 / ivf_flat.cc:786-800: trained, non-empty, and the
+    // raft log has run ahead of the last snapshot by > need_save_count
+    // (DEFINE_int64(..._need_save_count, 10000) in all three indexes)
+    dg_stats_out s{};
+    if (dg_stats(idx_, &s) != DG_OK) return false;
+    if (!s.is_trained || s.ntotal == 0) return false;
+    return last_save_log_behind > kNeedSaveCount;
+  }
+  bool SupportSave() override { return true; }
+  void LockWrite() override { dg_lock_write(idx_); }
+  void UnlockWrite() override { dg_unlock_write(idx_); }
   Status Save(const std::string& path) override {
-    return from_dg(dg_save(idx_, path.c_str()));
+    // the snapshot cycle ships faiss::write_index files
+    // (vector_index_snapshot_manager.cc:583-599) — emit the compatible
+    // container so CPU nodes can ingest the snapshot
+    return from_dg(dg_save_faiss(idx_, path.c_str()));
   }
   Status Load(const std::string& path) override {
     dg_index* ni = nullptr;
-    dg_status st = dg_load(&ni, path.c_str(), -1);
+    dg_status st = dg_load_faiss(&ni, path.c_str(), (int32_t)metric_, -1);
     if (st != DG_OK) return from_dg(st);
     dg_index_destroy(idx_);
     idx_ = ni;
@@ -236,10 +294,13 @@ class GpuIndexBase : public VectorIndex {
   }
 
  protected:
+  static constexpr int64_t kNeedSaveCount = 10000;  // *_need_save_count
+  static constexpr int64_t kMaxPointsPerCentroid = 256;  // faiss Clustering
   dg_index* idx_ = nullptr;
   dg_status create_st_ = DG_OK;
   MetricType metric_;
   int32_t dim_;
+  int64_t train_data_size_ = 0;
 };
 
 class GpuFlatIndex : public GpuIndexBase {
@@ -247,13 +308,32 @@ class GpuFlatIndex : public GpuIndexBase {
   GpuFlatIndex(MetricType m, int32_t d, int dev)
       : GpuIndexBase(DG_INDEX_FLAT, m, d, 0, dev) {}
   bool NeedTrain() override { return false; }
+  // Flat has no NeedToRebuild trigger in the reference (base default false)
 };
 
 class GpuIvfFlatIndex : public GpuIndexBase {
  public:
   GpuIvfFlatIndex(MetricType m, int32_t d, int32_t nlist, int dev)
-      : GpuIndexBase(DG_INDEX_IVF_FLAT, m, d, nlist, dev) {}
+      : GpuIndexBase(DG_INDEX_IVF_FLAT, m, d, nlist, dev),
+        nlist_org_(nlist) {}
   bool NeedTrain() override { return !IsTrained(); }
+  bool NeedToRebuild() override {
+    // restates vector_index_ivf_flat.cc:751-783: rebuild when the data has
+    // outgrown the structure (degraded nlist, or the train sample is now
+    // under half the data) past max_points_per_centroid * nlist
+    dg_stats_out s{};
+    if (dg_stats(idx_, &s) != DG_OK || !s.is_trained) return false;
+    const int64_t nlist_now = s.nlist;  // degrades to 1 at small trains
+    if (nlist_now == nlist_org_ && nlist_now == 1) return false;
+    const bool grown = s.ntotal >= kMaxPointsPerCentroid * nlist_org_;
+    if (nlist_now != nlist_org_ && nlist_now == 1 && grown) return true;
+    if (nlist_now == nlist_org_ && nlist_now != 1 && grown)
+      return train_data_size_ <= s.ntotal / 2;
+    return false;
+  }
+
+ private:
+  int64_t nlist_org_;
 };
 
 class GpuIvfPqIndex : public GpuIndexBase {
@@ -262,6 +342,12 @@ class GpuIvfPqIndex : public GpuIndexBase {
                 int dev)
       : GpuIndexBase(DG_INDEX_IVF_PQ, m, d, nlist, dev, pq_m) {}
   bool NeedTrain() override { return !IsTrained(); }
+  bool NeedToRebuild() override {
+    // vector_index_raw_ivf_pq.cc:518-526
+    dg_stats_out s{};
+    if (dg_stats(idx_, &s) != DG_OK || !s.is_trained) return false;
+    return (s.ntotal / 2) >= train_data_size_;
+  }
 };
 
 }  // namespace
@@ -383,6 +469,8 @@ extern "C" int dg_mirror_selftest(void) {
     if (!has_self) return 32;
     // delete + upsert
     if (!ivf->Delete({7}).ok()) return 33;
+    int64_t delc = -1;
+    if (!ivf->GetDeletedCount(delc).ok() || delc != 1) return 61;
     std::vector<VectorWithDistanceResult> r3;
     ivf->Search({batch[7]}, 1, {}, false, p, r3);
     if (!r3[0].vector_with_distances.empty() &&
@@ -409,6 +497,56 @@ extern "C" int dg_mirror_selftest(void) {
     bad.vector.dimension = d2 / 2;
     bad.vector.float_values.resize(d2 / 2);
     if (ivf->Add({bad}).ok()) return 40;
+
+    // ---- wrapper-lifecycle virtuals (§8b must-implement set) ----
+    if (!ivf->SupportSave()) return 41;
+    if (ivf->IsExceedsMaxElements(1 << 20)) return 42;  // always false
+    int64_t delc2 = -1;  // faiss-file load dropped the tombstones
+    if (!ivf->GetDeletedCount(delc2).ok() || delc2 != 0) return 43;
+    if (ivf->NeedToSave(100)) return 44;      // behind <= need_save_count
+    if (!ivf->NeedToSave(20000)) return 45;   // behind > 10000
+    if (ivf->NeedToRebuild()) return 46;      // ntotal < 256 * nlist
+    ivf->LockWrite();   // exclusive fork-save window lock
+    ivf->UnlockWrite();
+    // raw-float Train(std::vector<float>&) on a fresh index
+    {
+      auto ivf2 = NewIvfFlatIndex(MetricType::kL2, d2, 4);
+      if (!ivf2) return 47;
+      std::vector<float> tdata;
+      tdata.reserve((size_t)n2 * d2);
+      for (auto& vw : batch)
+        tdata.insert(tdata.end(), vw.vector.float_values.begin(),
+                     vw.vector.float_values.end());
+      std::vector<float> badsize(tdata.begin(), tdata.begin() + d2 / 2);
+      if (ivf2->Train(badsize).ok()) return 48;  // size % dim != 0 rejected
+      if (!ivf2->Train(tdata).ok()) return 49;
+      if (!ivf2->IsTrained()) return 50;
+    }
+    // ConcreteFilterFunctor (IDSelectorBatch semantics) incl. Build hook
+    {
+      std::vector<int64_t> want{3, 9, 15};
+      auto cf = std::make_shared<ConcreteFilterFunctor>(want);
+      std::vector<int64_t> idmap;
+      cf->Build(idmap);  // reference default no-op hook, callable
+      if (!cf->Check(9) || cf->Check(10)) return 51;
+      std::vector<std::shared_ptr<FilterFunctor>> fs{cf};
+      std::vector<VectorWithDistanceResult> r6;
+      if (!ivf->Search({batch[3]}, 3, fs, false, p, r6).ok()) return 52;
+      if (r6[0].vector_with_distances.empty()) return 53;
+      for (auto& vd : r6[0].vector_with_distances) {
+        int64_t id = vd.vector_with_id.id;
+        if (id != 3 && id != 9 && id != 15) return 54;
+      }
+      // negated form excludes them
+      auto nf = std::make_shared<ConcreteFilterFunctor>(want, true);
+      std::vector<std::shared_ptr<FilterFunctor>> fs2{nf};
+      std::vector<VectorWithDistanceResult> r7;
+      if (!ivf->Search({batch[3]}, 5, fs2, false, p, r7).ok()) return 55;
+      for (auto& vd : r7[0].vector_with_distances) {
+        int64_t id = vd.vector_with_id.id;
+        if (id == 3 || id == 9 || id == 15) return 56;
+      }
+    }
   }
   return 0;
 }

@@ -75,6 +75,9 @@ constexpr int kEInternal = 10002;
 class FilterFunctor {
  public:
   virtual ~FilterFunctor() = default;
+  // hook the reference gives filters to pre-index the id universe
+  // (vector_index.h:70); default no-op like the reference
+  virtual void Build(std::vector<int64_t>& /*id_map*/) {}
   virtual bool Check(int64_t vector_id) = 0;
   // translate to the device filter; default: unsupported (caller falls back
   // to post-filtering like the reader's over-fetch path)
@@ -117,20 +120,55 @@ class SortFilterFunctor : public FilterFunctor {  // vector_index.h:109-146
   bool negation_;
 };
 
-// ---- VectorIndex mirror (vector_index.h:148-229 virtuals) ----
+// ConcreteFilterFunctor mirror (vector_index.h:86-106: faiss IDSelectorBatch
+// membership + optional negation); device form = sorted-id filter
+class ConcreteFilterFunctor : public FilterFunctor {
+ public:
+  explicit ConcreteFilterFunctor(const std::vector<int64_t>& ids,
+                                 bool is_negation = false);
+  bool Check(int64_t id) override;
+  bool ToDeviceFilter(dg_filter* out) const override {
+    *out = {};
+    out->kind = DG_FILTER_SORTED_IDS;
+    out->ids = sorted_.data();
+    out->n_ids = (int64_t)sorted_.size();
+    out->negate = negation_ ? 1 : 0;
+    return true;
+  }
+
+ private:
+  std::vector<int64_t> sorted_;
+  bool negation_;
+};
+
+// ---- VectorIndex mirror (vector_index.h:148-229 virtuals; every
+// must-implement virtual the wrapper/manager calls is present) ----
 class VectorIndex {
  public:
   virtual ~VectorIndex() = default;
   virtual int32_t GetDimension() = 0;
   virtual MetricType GetMetricType() = 0;
   virtual Status GetCount(int64_t& count) = 0;
+  virtual Status GetDeletedCount(int64_t& deleted_count) = 0;
   virtual Status GetMemorySize(int64_t& bytes) = 0;
+  virtual bool IsExceedsMaxElements(int64_t vector_size) = 0;
   virtual Status Add(const std::vector<VectorWithId>& v) = 0;
   virtual Status Upsert(const std::vector<VectorWithId>& v) = 0;
   virtual Status Delete(const std::vector<int64_t>& ids) = 0;
   virtual Status Train(const std::vector<VectorWithId>& v) = 0;
+  // raw float form (vector_index.h:196: Train(std::vector<float>&))
+  virtual Status Train(std::vector<float>& train_datas) = 0;
   virtual bool IsTrained() = 0;
   virtual bool NeedTrain() = 0;
+  // rebuild/save scheduling hooks VectorIndexManager drives
+  // (vector_index.h:198-204; semantics per concrete index)
+  virtual bool NeedToRebuild() = 0;
+  virtual bool NeedToSave(int64_t last_save_log_behind) = 0;
+  virtual bool SupportSave() { return false; }
+  // exclusive lock for the wrapper's fork-save window
+  // (vector_index.h:192-193; rw_lock_.LockWrite/UnlockWrite)
+  virtual void LockWrite() = 0;
+  virtual void UnlockWrite() = 0;
   virtual Status Save(const std::string& path) = 0;
   virtual Status Load(const std::string& path) = 0;
   virtual Status Search(const std::vector<VectorWithId>& queries,

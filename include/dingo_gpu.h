@@ -126,6 +126,10 @@ typedef struct dg_stats_out {
    * (grouped accounting, SURVEY.md §8d cfg C) */
   int64_t last_scan_bytes_algorithmic;
   double last_scan_gbps_algorithmic;
+  /* tombstoned-but-uncompacted rows (GetDeletedCount, vector_index.h:151;
+   * the reference's faiss path compacts on remove so it reports 0 — here
+   * removes tombstone until the next finalize) */
+  int64_t deleted_count;
 } dg_stats_out;
 
 /* ---- lifecycle ---- */
@@ -234,6 +238,12 @@ dg_status dg_load_faiss(dg_index** out, const char* path,
  * the caller.  An optional list mask restricts scanning to owned lists for
  * list-sharded deployments: mask[l] != 0 => list l is scanned here. */
 dg_status dg_set_list_mask(dg_index* idx, const uint8_t* mask /* nlist */);
+
+/* ---- wrapper-lifecycle lock (LockWrite/UnlockWrite,
+ * vector_index.h:192-193): the exclusive lock the reference wrapper takes
+ * around its fork-save window; must be released by the same thread. ---- */
+void dg_lock_write(dg_index* idx);
+void dg_unlock_write(dg_index* idx);
 
 /* ---- introspection ---- */
 dg_status dg_stats(dg_index* idx, dg_stats_out* out);
