@@ -185,10 +185,16 @@ def compute_recall(idx, q, k, nprobe, nlist, world, device):
     return hits / (nq * k)
 
 
-def cpu_baseline(idx, args, q_host):
+def cpu_baseline(idx, args, q_host, gpu_ids=None):
     """Oracle (kind 'port') timed on host cores, same structure: centroids
     from the GPU index, member lists from its assignments, base regenerated
-    on host from the same protocol. Bounded sample (~10-30 s)."""
+    on host from the same protocol. Bounded sample (~10-30 s).
+
+    When gpu_ids (the GPU path's [nq, k] result ids on the same queries) is
+    given and the oracle runs over the FULL database, also reports
+    recall_cross_engine: mean top-k overlap of GPU vs oracle result sets at
+    the bench scale (VERDICT r01 weak 1 — closes the self-referential
+    recall gap)."""
     sys.path.insert(0, os.path.join(REPO, "oracle"))
     import pyoracle as orc
     import psutil
@@ -224,15 +230,23 @@ def cpu_baseline(idx, args, q_host):
     per_q = (time.time() - t0) / nq_probe
     nq_sample = int(min(len(q_host), max(8, 20.0 / max(per_q, 1e-6))))
     t0 = time.time()
-    orc.ivf_search_indexed_fast(orc.L2, cents, offsets, member_rows, base,
-                                q_host[:nq_sample], k, nprobe)
+    o_dist, o_ids = orc.ivf_search_indexed_fast(
+        orc.L2, cents, offsets, member_rows, base, q_host[:nq_sample], k,
+        nprobe)
     dt = time.time() - t0
     cores = os.cpu_count()
     sample = (f"{nq_sample} queries over "
               f"{'full' if frac == 1.0 else f'{frac:.0%}-row-subsampled'} "
               f"database ({n_cpu} rows), {dt:.1f}s, OpenMP {cores} cores")
-    return {"value": round(nq_sample / dt, 2), "unit": "queries/s",
-            "cores": cores, "kind": "port", "sample": sample}
+    out = {"value": round(nq_sample / dt, 2), "unit": "queries/s",
+           "cores": cores, "kind": "port", "sample": sample}
+    if gpu_ids is not None and frac == 1.0:
+        nq_x = min(len(o_ids), len(gpu_ids), 256)
+        hits = sum(len(set(gpu_ids[r]) & set(o_ids[r]))
+                   for r in range(nq_x))
+        out["recall_cross_engine"] = round(hits / (nq_x * k), 4)
+        out["recall_cross_engine_nq"] = nq_x
+    return out
 
 
 def cpu_baseline_flat(idx, args, q_host):
@@ -296,6 +310,10 @@ def main():
                     help="reduced size for smoke runs (1M rows)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--no-recall", action="store_true")
+    ap.add_argument("--recall-curve", action="store_true",
+                    help="also sweep nprobe in {32,64,128,256} and write the"
+                         " QPS-vs-recall operating curve to"
+                         " gpurun_out/recall_curve.json")
     args = ap.parse_args()
     if args.quick:
         args.n, args.nlist = 1_000_000, 1024
@@ -365,9 +383,13 @@ def main():
             and args.kind in ("ivf_flat", "flat")):
         q_host = q.cpu().numpy()
         try:
-            cpu = (cpu_baseline_flat(idx, args, q_host)
-                   if args.kind == "flat" else
-                   cpu_baseline(idx, args, q_host))
+            if args.kind == "flat":
+                cpu = cpu_baseline_flat(idx, args, q_host)
+            else:
+                # fresh GPU result at the bench nprobe for the cross-engine
+                # check (compute_recall left the exact sweep in ids_t)
+                merged_step(idx, q, k, nprobe, dist_t, ids_t, world, bufs)
+                cpu = cpu_baseline(idx, args, q_host, ids_t.cpu().numpy())
         except Exception as ex:
             log(rank, f"cpu baseline failed: {ex}")
 
@@ -413,6 +435,10 @@ def main():
             "data": "synthetic (seeded torch philox uniform[0,1); queries = "
                     "base + N(0,0.05); DESIGN.md §data)",
             "recall_at_k": round(recall, 4) if recall is not None else None,
+            # GPU top-k set overlap vs the CPU oracle on the same structure
+            # and queries at full bench scale (null when the oracle leg was
+            # skipped or row-subsampled)
+            "recall_cross_engine": (cpu or {}).get("recall_cross_engine"),
             "config": {
                 "workload": workload_name,
                 "n": args.n, "d": args.d, "nlist": args.nlist,
@@ -439,6 +465,35 @@ def main():
             "cpu_baseline": cpu,
         }
         print(json.dumps(out), flush=True)
+
+    if (args.recall_curve and rank == 0 and world == 1
+            and args.kind == "ivf_flat"):
+        # QPS-vs-recall operating curve: same index,
+        # nprobe swept; recall vs the full sweep as in compute_recall
+        curve = []
+        for np_i in (32, 64, 128, 256):
+            if np_i > args.nlist:
+                continue
+            for _ in range(2):
+                merged_step(idx, q, k, np_i, dist_t, ids_t, world, bufs)
+            torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(5):
+                merged_step(idx, q, k, np_i, dist_t, ids_t, world, bufs)
+            torch.cuda.synchronize()
+            dt = time.time() - t0
+            rec = compute_recall(idx, q, k, np_i, args.nlist, world, device)
+            curve.append({"nprobe": np_i, "qps": round(nq * 5 / dt, 1),
+                          "ms_per_step": round(dt / 5 * 1000, 3),
+                          "recall_at_k": round(rec, 4)})
+            log(rank, f"curve nprobe={np_i}: {curve[-1]}")
+        os.makedirs(os.path.join(REPO, "gpurun_out"), exist_ok=True)
+        json.dump({"workload": "cfg C structure", "n": args.n, "d": args.d,
+                   "nlist": args.nlist, "batch": nq, "k": k,
+                   "curve": curve},
+                  open(os.path.join(REPO, "gpurun_out",
+                                    "recall_curve.json"), "w"), indent=1)
+
     idx.close()
     if world > 1:
         torch.distributed.destroy_process_group()
