@@ -1107,6 +1107,236 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
   }
 }
 
+// ---------- fused-stage asm scan (v4, DG_SCAN_VARIANT=13) ----------
+// The v3 kernel's separate issue/wait asms are not airtight: LLVM may copy
+// or reuse an in-flight load's destination registers between the asm
+// statements (it models asm loads as completing instantly; verified in the
+// emitted code by a static hazard scan).  Here each 8-dim stage is ONE asm
+// block — 8 global_load_dwordx4 into the next bank, address bumps, and the
+// counted s_waitcnt vmcnt(8) that retires the CURRENT bank — so no
+// compiler-scheduled instruction can fall between a load and the wait that
+// guards its window.  Two banks ping-pong; the current bank's values are
+// tied through ("+v") so their consumers order after the wait.
+#define DG_STAGE8(c0, c1, c2, c3, c4, c5, c6, c7, n0, n1, n2, n3, n4, n5,  \
+                  n6, n7, addr, stride)                                    \
+  asm volatile("global_load_dwordx4 %9, %8, off\n\t"                       \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %10, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %11, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %12, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %13, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %14, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %15, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "global_load_dwordx4 %16, %8, off\n\t"                      \
+               "v_lshl_add_u64 %8, %8, 0, %17\n\t"                         \
+               "s_waitcnt vmcnt(8)"                                        \
+               : "+v"(c0), "+v"(c1), "+v"(c2), "+v"(c3), "+v"(c4),         \
+                 "+v"(c5), "+v"(c6), "+v"(c7), "+v"(addr), "=&v"(n0),      \
+                 "=&v"(n1), "=&v"(n2), "=&v"(n3), "=&v"(n4), "=&v"(n5),    \
+                 "=&v"(n6), "=&v"(n7)                                      \
+               : "v"(stride))
+
+// prologue: issue one bank, no wait
+#define DG_ISSUE8(n0, n1, n2, n3, n4, n5, n6, n7, addr, stride)            \
+  asm volatile("global_load_dwordx4 %1, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %2, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %3, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %4, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %5, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %6, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %7, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9\n\t"                          \
+               "global_load_dwordx4 %8, %0, off\n\t"                       \
+               "v_lshl_add_u64 %0, %0, 0, %9"                              \
+               : "+v"(addr), "=&v"(n0), "=&v"(n1), "=&v"(n2), "=&v"(n3),   \
+                 "=&v"(n4), "=&v"(n5), "=&v"(n6), "=&v"(n7)                \
+               : "v"(stride))
+
+// wait draining everything, tying one bank's consumers behind it
+#define DG_DRAIN8(b0, b1, b2, b3, b4, b5, b6, b7)                          \
+  asm volatile("s_waitcnt vmcnt(0)"                                        \
+               : "+v"(b0), "+v"(b1), "+v"(b2), "+v"(b3), "+v"(b4),         \
+                 "+v"(b5), "+v"(b6), "+v"(b7))
+
+template <int QTM>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe2(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  constexpr int RPL = 4;
+  extern __shared__ __attribute__((aligned(16))) float smem[];  // [QTM * d]
+  int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);          // [QTM]
+
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* col = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+  const int64_t row0 = list_start + (int64_t)chunk * chunk_rows;
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  __builtin_assume(d % 4 == 0 && d > 0);
+
+  for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
+    const int32_t qt = min(QTM, nql - t0);
+    __syncthreads();
+    for (int32_t j = 0; j < qt; j++) {
+      int32_t q = inv_q[iq0 + t0 + j];
+      const float4* src = (const float4*)(queries + (size_t)q * d);
+      float4* dst = (float4*)(smem + (size_t)j * d);
+      for (int i = threadIdx.x; i < d / 4; i += blockDim.x) dst[i] = src[i];
+    }
+    for (size_t i = (size_t)qt * d + threadIdx.x; i < (size_t)QTM * d;
+         i += blockDim.x)
+      smem[i] = 0.f;
+    if (threadIdx.x < QTM) {
+      int32_t j = threadIdx.x;
+      if (j < qt) {
+        int32_t q = inv_q[iq0 + t0 + j];
+        int32_t rank = inv_rank[iq0 + t0 + j];
+        cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                   list_start;
+      } else {
+        cbase[j] = 0;
+      }
+    }
+    __syncthreads();
+
+    for (int32_t rb = wave_id * WAVE * RPL; rb < nrows_pad;
+         rb += 4 * WAVE * RPL) {
+      const int32_t rr0 = rb + lane * RPL;
+      if (rr0 >= nrows_pad) continue;
+      float acc[QTM][RPL];
+#pragma unroll
+      for (int j = 0; j < QTM; j++)
+#pragma unroll
+        for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
+
+      // drain previous-tile vmem so the fused counts are exact
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+      const size_t cstride = (size_t)nrows_pad * 4;
+      uint64_t addr = (uint64_t)(uintptr_t)(col + rr0);
+      const uint64_t stride = cstride;
+      dg_f4 a0, a1, a2, a3, a4, a5, a6, a7;
+      dg_f4 b0, b1, b2, b3, b4, b5, b6, b7;
+
+      // 8 dims of one bank against all staged queries.  The sched barrier
+      // every 4 queries stops the scheduler from hoisting ALL queries' LDS
+      // reads at once (measured: 96 live query registers, pushing the
+      // allocator into AGPR stashes of the in-flight bank = reads of
+      // pending load destinations).
+      auto compute8 = [&](const dg_f4& x0, const dg_f4& x1, const dg_f4& x2,
+                          const dg_f4& x3, const dg_f4& x4, const dg_f4& x5,
+                          const dg_f4& x6, const dg_f4& x7, int32_t ib) {
+        const dg_f4* xs[8] = {&x0, &x1, &x2, &x3, &x4, &x5, &x6, &x7};
+#pragma unroll
+        for (int j = 0; j < QTM; j++) {
+          const float* qp = smem + (size_t)j * d + ib;
+#pragma unroll
+          for (int h = 0; h < 4; h++) {
+            const float2 qq = *(const float2*)(qp + 2 * h);
+            const dg_f4& ca = *xs[2 * h];
+            const dg_f4& cb = *xs[2 * h + 1];
+#pragma unroll
+            for (int x = 0; x < RPL; x++)
+              acc[j][x] += ca[x] * qq.x + cb[x] * qq.y;
+          }
+          if ((j & 3) == 3) __builtin_amdgcn_sched_barrier(0);
+        }
+      };
+
+      int32_t base = 0;
+      if (d >= 16) {
+        DG_ISSUE8(a0, a1, a2, a3, a4, a5, a6, a7, addr, stride);
+        for (; base + 16 <= d; base += 16) {
+          // sched barriers pin each compute phase between its stage asms:
+          // without them the scheduler sinks FMA chunks past the next
+          // stage, doubling bank liveness and pushing the allocator into
+          // AGPR stashes of in-flight destinations
+          DG_STAGE8(a0, a1, a2, a3, a4, a5, a6, a7, b0, b1, b2, b3, b4, b5,
+                    b6, b7, addr, stride);
+          __builtin_amdgcn_sched_barrier(0);
+          compute8(a0, a1, a2, a3, a4, a5, a6, a7, base);
+          __builtin_amdgcn_sched_barrier(0);
+          DG_STAGE8(b0, b1, b2, b3, b4, b5, b6, b7, a0, a1, a2, a3, a4, a5,
+                    a6, a7, addr, stride);
+          __builtin_amdgcn_sched_barrier(0);
+          compute8(b0, b1, b2, b3, b4, b5, b6, b7, base + 8);
+          __builtin_amdgcn_sched_barrier(0);
+        }
+        DG_DRAIN8(a0, a1, a2, a3, a4, a5, a6, a7);
+      }
+      // tail (< 16 dims): compiler-managed loads
+      const float* colp = col + rr0;
+      for (int32_t ib = base; ib < d; ib += 4) {
+        float4 c[4];
+#pragma unroll
+        for (int u = 0; u < 4; u++)
+          c[u] = (ib + u < d) ? *(const float4*)(colp +
+                                                 (size_t)(ib + u) * nrows_pad)
+                              : float4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int j = 0; j < QTM; j++) {
+          const float* qp = smem + (size_t)j * d + ib;
+#pragma unroll
+          for (int u = 0; u < 4 && ib + u < d; u++) {
+            const float qv = qp[u];
+            acc[j][0] += c[u].x * qv;
+            acc[j][1] += c[u].y * qv;
+            acc[j][2] += c[u].z * qv;
+            acc[j][3] += c[u].w * qv;
+          }
+        }
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+#pragma unroll
+      for (int j = 0; j < QTM; j++) {
+        if (j < qt) {
+          const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+          for (int x = 0; x < RPL; x++) {
+            const int32_t rl = rr0 + x;
+            if (rl < nrows) {
+              const int64_t r = row0 + rl;
+              bool pass = true;
+              if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+              float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                        : -acc[j][x];
+              cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
 // ---------- glds scan (v3 experiment, DG_SCAN_VARIANT=7) ----------
 // Column data streams HBM -> LDS via global_load_lds into a per-wave
 // private ring (each wave owns a 256-row quarter of the chunk, so no
@@ -1789,7 +2019,32 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                          metric, bitmap, chunk_rows, cand);
       break;
     }
-    case 12: {  // asm-pipelined, full-drain waits (correctness baseline)
+    case 13: {  // fused-stage asm scan (v4)
+      constexpr int QTM = 16;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_pipe2<QTM>), dim3((uint32_t)n_units),
+                         dim3(256), lds, s, units, csr_offsets, chunk_off,
+                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
+                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
+                         metric, bitmap, chunk_rows, cand);
+      break;
+    }
+    case 14: {  // fused-stage asm scan, QTM=12 (lower register pressure)
+      constexpr int QTM = 12;
+      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
+      hipLaunchKernelGGL((k_ivf_scan_pipe2<QTM>), dim3((uint32_t)n_units),
+                         dim3(256), lds, s, units, csr_offsets, chunk_off,
+                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
+                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
+                         metric, bitmap, chunk_rows, cand);
+      break;
+    }
+    case 0:
+    case 12: {  // DEFAULT (round 2): asm-load pipeline with full-drain
+      // waits — the counted-wait variants (10/13/14) are faster on paper
+      // but LLVM's allocator cannot be made to respect in-flight asm-load
+      // destinations (verified by static hazard scan + GPU faults); the
+      // drain form is airtight and measured fastest of the correct set.
       constexpr int QTM = 16;
       size_t lds = (size_t)QTM * d * 4 + QTM * 8;
       hipLaunchKernelGGL((k_ivf_scan_pipe<QTM, true>),
@@ -1798,6 +2053,10 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                          queries, d, inv_offsets, inv_q, inv_rank, qp_off,
                          q_cand_base, nprobe, metric, bitmap, chunk_rows,
                          cand);
+      break;
+    }
+    case 15: {  // round-1 columnar pipeline (previous default)
+      DG_SCAN_LAUNCH(16, 4);
       break;
     }
     case 11: {  // asm-pipelined, QTM=12
