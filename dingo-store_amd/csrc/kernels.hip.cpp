@@ -2172,15 +2172,17 @@ void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
   // whose codes + S_l fit LDS wins unless occupancy dominates (A/B'd;
   // DG_PQ_RPV forces a tile for experiments).  TILE = RPV * 64 rows.
   const char* e = getenv("DG_PQ_RPV");
-  int rpv = e ? atoi(e) : 0;
-  if (rpv != 4 && rpv != 8 && rpv != 16) {
-    rpv = 16;
-    while (rpv > 4 &&
-           (size_t)rpv * 64 * M + (size_t)M * 256 * 2 > (158u << 10))
-      rpv >>= 1;
-  }
-  size_t lds = (size_t)rpv * 64 * M + (size_t)M * 256 * 2;  // codes + f16 S_l
-  if (rpv == 16)
+  int rpv = e ? atoi(e) : 4;  // 256-row tiles + 2 blocks/CU measured best
+  if (rpv != 4 && rpv != 6 && rpv != 8 && rpv != 16) rpv = 4;
+  size_t lds = (size_t)(rpv == 6 ? 4 : rpv) * 64 * M +
+               (size_t)M * 256 * 2;  // codes + f16 S_l
+  if (rpv == 6)  // round-1 form: VGPRs capped at 80 by launch_bounds(256,6)
+    hipLaunchKernelGGL((k_ivfpq_scan<4, 6>), dim3((uint32_t)n_units),
+                       dim3(256), lds, s, units, csr_offsets, csr_codes, S,
+                       T, coarse_dots, nlist, M, inv_offsets, inv_q,
+                       inv_rank, qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
+  else if (rpv == 16)
     hipLaunchKernelGGL((k_ivfpq_scan<16, 1>), dim3((uint32_t)n_units), dim3(256),
                        lds, s, units, csr_offsets, csr_codes, S, T,
                        coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
@@ -2192,11 +2194,12 @@ void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
                        coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
                        qp_off, q_cand_base, nprobe, metric, bitmap,
                        chunk_rows, cand);
-  else
-    hipLaunchKernelGGL((k_ivfpq_scan<4, 6>), dim3((uint32_t)n_units), dim3(256),
-                       lds, s, units, csr_offsets, csr_codes, S, T,
-                       coarse_dots, nlist, M, inv_offsets, inv_q, inv_rank,
-                       qp_off, q_cand_base, nprobe, metric, bitmap,
+  else  // default: 256-row tiles with the register cap lifted (2 blocks/CU
+        // by LDS anyway; 80-VGPR cap limited outstanding T gathers)
+    hipLaunchKernelGGL((k_ivfpq_scan<4, 2>), dim3((uint32_t)n_units),
+                       dim3(256), lds, s, units, csr_offsets, csr_codes, S,
+                       T, coarse_dots, nlist, M, inv_offsets, inv_q,
+                       inv_rank, qp_off, q_cand_base, nprobe, metric, bitmap,
                        chunk_rows, cand);
 }
 

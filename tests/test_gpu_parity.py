@@ -162,6 +162,50 @@ def test_ivf_recall_equals_oracle():
     assert abs(r_gpu - r_orc) <= 0.001, (r_gpu, r_orc)  # ±0.1pp, §8c
 
 
+def test_wide_k_flat():
+    # round-1 capped k at 128; the segmented selector lifts it to 2048
+    # (reference accepts arbitrary top_n, conf vector_max_batch_count=4096)
+    base, q = make_data(n=6000, d=64, nq=8)
+    k = 300
+    gpu = dg.Index(dg.FLAT, dg.L2, 64)
+    gpu.add(np.arange(6000, dtype=np.int64), base)
+    try:
+        gd, gi = gpu.search(q, k)
+    finally:
+        gpu.close()
+    od, oi = orc.flat_search(orc.L2, base, q, k)
+    frac = ids_match_with_tie_slack(gd, gi, od, oi)
+    assert frac >= 0.98, frac
+
+
+def test_wide_k_and_nprobe_ivf():
+    # nprobe > 128 (and < nlist) + k > 128 together, vs the oracle
+    base, q = make_data(n=30000, d=64, nq=16)
+    nlist, nprobe, k = 512, 256, 200
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_, q, k, nprobe)
+    frac = ids_match_with_tie_slack(gd, gi, od, oi)
+    assert frac >= 0.98, frac
+
+
+def test_wide_k_1024_ivf():
+    # k = 1024 (wrapper-scale top_n) through the candidate selector
+    base, q = make_data(n=20000, d=64, nq=4)
+    nlist, nprobe, k = 64, 32, 1024
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        gd, gi = gpu.search(q, k, nprobe=nprobe)
+    finally:
+        gpu.close()
+    od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_, q, k, nprobe)
+    frac = ids_match_with_tie_slack(gd, gi, od, oi)
+    assert frac >= 0.98, frac
+
+
 def test_ivf_nprobe_default_and_clamp():
     # nprobe<=0 -> default 80 clamped to nlist (ivf_flat.cc:208-214,234)
     base, q = make_data(n=5000, d=64, nq=16)
