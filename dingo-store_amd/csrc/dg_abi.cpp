@@ -150,6 +150,41 @@ static void rand_perm(std::vector<int64_t>& perm, int64_t n, uint32_t seed) {
     std::swap(perm[i], perm[i + rng.rand_int(n - i)]);
 }
 
+// upload n rows of d_user-strided floats (host or device source) into a
+// PADDED device region (row stride ix->desc.d), zero-filling pad columns —
+// zeros leave dots, norms and cosine normalization unchanged
+static dg_status upload_rows_padded(dg_index* ix, float* dst,
+                                    const float* src, int64_t n,
+                                    bool src_on_device) {
+  const int32_t du = ix->d_user, dp = ix->desc.d;
+  const hipMemcpyKind kind =
+      src_on_device ? hipMemcpyDeviceToDevice : hipMemcpyHostToDevice;
+  if (du == dp) {
+    DG_HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)n * du * 4, kind,
+                                ix->stream));
+    return DG_OK;
+  }
+  DG_HIP_CHECK(hipMemsetAsync(dst, 0, (size_t)n * dp * 4, ix->stream));
+  DG_HIP_CHECK(hipMemcpy2DAsync(dst, (size_t)dp * 4, src, (size_t)du * 4,
+                                (size_t)du * 4, (size_t)n, kind, ix->stream));
+  return DG_OK;
+}
+
+// download padded device rows into a d_user-strided host buffer (sync)
+static dg_status download_rows_strip(dg_index* ix, float* dst,
+                                     const float* src, int64_t n) {
+  const int32_t du = ix->d_user, dp = ix->desc.d;
+  if (du == dp) {
+    DG_HIP_CHECK(hipMemcpy(dst, src, (size_t)n * du * 4,
+                           hipMemcpyDeviceToHost));
+    return DG_OK;
+  }
+  DG_HIP_CHECK(hipMemcpy2D(dst, (size_t)du * 4, src, (size_t)dp * 4,
+                           (size_t)du * 4, (size_t)n,
+                           hipMemcpyDeviceToHost));
+  return DG_OK;
+}
+
 // per-index exclusive-scan scratch (was a process-wide static
 // racing across indexes/devices).  n = scan length; scratch holds one i64
 // per 256-element block.
@@ -208,11 +243,12 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
     return DG_EINVAL;
   }
   dg_index_desc desc = *dp;
-  if (desc.d <= 0 || desc.d % 4 != 0 || desc.d > 8192) {
-    dg_set_error(
-        "dimension %d unsupported (need 0 < d <= 8192, d %% 4 == 0)", desc.d);
+  if (desc.d <= 0 || desc.d > 8192) {
+    dg_set_error("dimension %d unsupported (need 0 < d <= 8192)", desc.d);
     return desc.d <= 0 ? DG_EINVAL : DG_ENOT_SUPPORT;
   }
+  const int32_t d_user = desc.d;
+  desc.d = (desc.d + 3) & ~3;  // internal pad (see dg_internal.h)
   if (desc.metric < 0 || desc.metric > 2) {
     dg_set_error("bad metric %d", desc.metric);
     return DG_EINVAL;
@@ -224,9 +260,10 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
       dg_set_error("only nbits=8 supported (reference default, constant.h)");
       return DG_ENOT_SUPPORT;
     }
-    if (desc.d % desc.pq_m != 0 || desc.pq_m % 4 != 0) {
-      dg_set_error("need d %% m == 0 and m %% 4 == 0 (d=%d m=%d)", desc.d,
-                   desc.pq_m);
+    if (d_user % 4 != 0 || d_user % desc.pq_m != 0 || desc.pq_m % 4 != 0) {
+      dg_set_error(
+          "IVF-PQ needs d %% 4 == 0, d %% m == 0 and m %% 4 == 0 "
+          "(d=%d m=%d)", d_user, desc.pq_m);
       return DG_ENOT_SUPPORT;
     }
   }
@@ -237,16 +274,13 @@ extern "C" dg_status dg_index_create(dg_index** out, const dg_index_desc* dp) {
   }
   if (desc.kind != DG_INDEX_FLAT && desc.nlist <= 0)
     desc.nlist = 2048;  // kCreateIvfFlatParamNcentroids, constant.h:177
-  if (desc.kind == DG_INDEX_IVF_FLAT && desc.d > 2048) {
-    dg_set_error("IVF scan kernel supports d <= 2048 this round");
-    return DG_ENOT_SUPPORT;
-  }
   if (dg_device_count() == 0) {
     dg_set_error("no HIP device (the GPU path has no CPU fallback)");
     return DG_ENOGPU;
   }
   dg_index* ix = new dg_index();
   ix->desc = desc;
+  ix->d_user = d_user;
   ix->device = desc.device >= 0 ? desc.device : 0;
   DeviceGuard g(ix->device);
   if (hipStreamCreate(&ix->stream) != hipSuccess ||
@@ -323,8 +357,9 @@ extern "C" dg_status dg_set_centroids(dg_index* ix, int32_t nlist,
   size_t bytes = (size_t)nlist * ix->desc.d * 4;
   dg_status st = dbuf_reserve(ix->d_centroids, bytes, ix->stream, false);
   if (st != DG_OK) return st;
-  DG_HIP_CHECK(hipMemcpyAsync(ix->d_centroids.p, centroids, bytes,
-                              hipMemcpyHostToDevice, ix->stream));
+  st = upload_rows_padded(ix, (float*)ix->d_centroids.p, centroids, nlist,
+                          false);
+  if (st != DG_OK) return st;
   st = dbuf_reserve(ix->d_cnorms, (size_t)nlist * 4, ix->stream, false);
   if (st != DG_OK) return st;
   dgk::row_norms(ix->stream, (const float*)ix->d_centroids.p, nlist,
@@ -343,10 +378,8 @@ extern "C" dg_status dg_get_centroids(dg_index* ix, float* out) {
   }
   std::shared_lock lk(ix->rw);
   DeviceGuard g(ix->device);
-  DG_HIP_CHECK(hipMemcpy(out, ix->d_centroids.p,
-                         (size_t)ix->desc.nlist * ix->desc.d * 4,
-                         hipMemcpyDeviceToHost));
-  return DG_OK;
+  return download_rows_strip(ix, out, (const float*)ix->d_centroids.p,
+                             ix->desc.nlist);
 }
 
 // ---------------- train (GPU k-means; faiss Clustering semantics
@@ -588,11 +621,8 @@ extern "C" dg_status dg_train(dg_index* ix, int64_t n, const float* x) {
   dg_status st = dbuf_reserve(d_td, (size_t)n * d * 4, ix->stream, false);
   if (st != DG_OK) return st;
   do {
-    if (hipMemcpyAsync(d_td.p, x, (size_t)n * d * 4, hipMemcpyHostToDevice,
-                       ix->stream) != hipSuccess) {
-      st = DG_EINTERNAL;
+    if ((st = upload_rows_padded(ix, (float*)d_td.p, x, n, false)) != DG_OK)
       break;
-    }
     if (ix->desc.metric == DG_METRIC_COSINE)
       dgk::normalize_rows(ix->stream, (float*)d_td.p, n, d);
     if ((st = dbuf_reserve(ix->d_centroids, (size_t)nlist * d * 4, ix->stream,
@@ -780,10 +810,11 @@ static dg_status add_impl(dg_index* ix, int64_t n, const int64_t* ids,
   } else {
     dst = (float*)ix->d_vectors.p + (size_t)n0 * d;
   }
-  DG_HIP_CHECK(hipMemcpyAsync(
-      dst, x, (size_t)n * d * 4,
-      x_on_device ? hipMemcpyDeviceToDevice : hipMemcpyHostToDevice,
-      ix->stream));
+  st = upload_rows_padded(ix, dst, x, n, x_on_device);
+  if (st != DG_OK) {
+    dbuf_free(staging);
+    return st;
+  }
   DG_HIP_CHECK(hipMemcpyAsync((int64_t*)ix->d_ids.p + n0, ids, (size_t)n * 8,
                               hipMemcpyHostToDevice, ix->stream));
   if (ix->desc.metric == DG_METRIC_COSINE)
@@ -836,11 +867,11 @@ dg_status dg_ingest_rows(dg_index* ix, int64_t n, const int64_t* ids,
     st = dbuf_reserve(ix->d_codes, (size_t)(n0 + n) * ix->desc.pq_m,
                       ix->stream, true);
   if (st != DG_OK) return st;
-  if (!is_pq)
-    DG_HIP_CHECK(hipMemcpyAsync((float*)ix->d_vectors.p + (size_t)n0 * d, x,
-                                (size_t)n * d * 4, hipMemcpyHostToDevice,
-                                ix->stream));
-  else
+  if (!is_pq) {
+    st = upload_rows_padded(ix, (float*)ix->d_vectors.p + (size_t)n0 * d, x,
+                            n, false);
+    if (st != DG_OK) return st;
+  } else
     DG_HIP_CHECK(hipMemcpyAsync(
         (uint8_t*)ix->d_codes.p + (size_t)n0 * ix->desc.pq_m, codes,
         (size_t)n * ix->desc.pq_m, hipMemcpyHostToDevice, ix->stream));
@@ -1130,8 +1161,8 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
           DG_OK)
     return st;
   float* dq = (float*)ix->ws_queries.p;
-  DG_HIP_CHECK(hipMemcpyAsync(dq, d_x, (size_t)nq * d * 4,
-                              hipMemcpyDeviceToDevice, ix->stream));
+  st = upload_rows_padded(ix, dq, d_x, nq, true);
+  if (st != DG_OK) return st;
   if (metric == DG_METRIC_COSINE) dgk::normalize_rows(ix->stream, dq, nq, d);
   float* dqn = (float*)ix->ws_qnorms.p;
   if (metric == DG_METRIC_L2) dgk::row_norms(ix->stream, dq, nq, d, dqn);
@@ -1676,14 +1707,14 @@ extern "C" dg_status dg_search(dg_index* ix, int64_t nq, const float* x,
   auto& tls = g_tls_staging.for_device(ix->device);
   dg_dbuf &t_in = tls.in, &t_dist = tls.dist, &t_ids = tls.ids;
   dg_status st;
-  if ((st = dbuf_reserve(t_in, (size_t)nq * ix->desc.d * 4, ix->stream,
+  if ((st = dbuf_reserve(t_in, (size_t)nq * ix->d_user * 4, ix->stream,
                          false)) != DG_OK ||
       (st = dbuf_reserve(t_dist, (size_t)nq * k * 4, ix->stream, false)) !=
           DG_OK ||
       (st = dbuf_reserve(t_ids, (size_t)nq * k * 8, ix->stream, false)) !=
           DG_OK)
     return st;
-  DG_HIP_CHECK(hipMemcpyAsync(t_in.p, x, (size_t)nq * ix->desc.d * 4,
+  DG_HIP_CHECK(hipMemcpyAsync(t_in.p, x, (size_t)nq * ix->d_user * 4,
                               hipMemcpyHostToDevice, ix->stream));
   st = dg_search_device(ix, nq, (const float*)t_in.p, k, nprobe, filter,
                         (float*)t_dist.p, (int64_t*)t_ids.p);
@@ -1737,10 +1768,10 @@ extern "C" dg_status dg_range_search(dg_index* ix, int64_t nq, const float* x,
   std::shared_lock lk(ix->rw);
   dg_dbuf& t_in = g_tls_staging.for_device(ix->device).in;
   dg_status st;
-  if ((st = dbuf_reserve(t_in, (size_t)nq * ix->desc.d * 4, ix->stream,
+  if ((st = dbuf_reserve(t_in, (size_t)nq * ix->d_user * 4, ix->stream,
                          false)) != DG_OK)
     return st;
-  DG_HIP_CHECK(hipMemcpyAsync(t_in.p, x, (size_t)nq * ix->desc.d * 4,
+  DG_HIP_CHECK(hipMemcpyAsync(t_in.p, x, (size_t)nq * ix->d_user * 4,
                               hipMemcpyHostToDevice, ix->stream));
   dg_range_req rr{radius, lims, out_ids, out_dists};
   // nprobe: reference RangeSearch uses the index default (clamped); pass 0
@@ -1766,16 +1797,19 @@ extern "C" dg_status dg_save(dg_index* ix, const char* path) {
   const int32_t d = ix->desc.d;
   int32_t trained = ix->trained ? 1 : 0;
   fwrite(&kMagic, 4, 1, f);
-  fwrite(&ix->desc, sizeof(ix->desc), 1, f);
+  dg_index_desc udesc = ix->desc;
+  udesc.d = ix->d_user;  // container speaks the caller's dimension
+  fwrite(&udesc, sizeof(udesc), 1, f);
   fwrite(&trained, 4, 1, f);
   fwrite(&ix->ntotal, 8, 1, f);
   fwrite(&ix->n_deleted, 8, 1, f);
   dg_status st = DG_OK;
   const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
   if (ix->desc.kind != DG_INDEX_FLAT && ix->trained) {
-    std::vector<float> cents((size_t)ix->desc.nlist * d);
-    if (hipMemcpy(cents.data(), ix->d_centroids.p, cents.size() * 4,
-                  hipMemcpyDeviceToHost) != hipSuccess)
+    std::vector<float> cents((size_t)ix->desc.nlist * ix->d_user);
+    if (download_rows_strip(ix, cents.data(),
+                            (const float*)ix->d_centroids.p,
+                            ix->desc.nlist) != DG_OK)
       st = DG_EINTERNAL;
     fwrite(cents.data(), 4, cents.size(), f);
   }
@@ -1790,7 +1824,7 @@ extern "C" dg_status dg_save(dg_index* ix, const char* path) {
   if (st == DG_OK && ix->ntotal > 0) {
     const size_t CH = 1 << 20;  // rows per host staging chunk
     const int32_t M = ix->desc.pq_m;
-    std::vector<float> vbuf(is_pq ? 0 : CH * d);
+    std::vector<float> vbuf(is_pq ? 0 : CH * ix->d_user);
     std::vector<uint8_t> cbuf(is_pq ? CH * M : 0);
     std::vector<int64_t> ibuf(CH);
     std::vector<int32_t> abuf(CH);
@@ -1802,10 +1836,11 @@ extern "C" dg_status dg_save(dg_index* ix, const char* path) {
           st = DG_EINTERNAL;
         fwrite(cbuf.data(), 1, c * M, f);
       } else {
-        if (hipMemcpy(vbuf.data(), (float*)ix->d_vectors.p + (size_t)s0 * d,
-                      c * d * 4, hipMemcpyDeviceToHost) != hipSuccess)
+        if (download_rows_strip(ix, vbuf.data(),
+                                (float*)ix->d_vectors.p + (size_t)s0 * d,
+                                (int64_t)c) != DG_OK)
           st = DG_EINTERNAL;
-        fwrite(vbuf.data(), 4, c * d, f);
+        fwrite(vbuf.data(), 4, c * ix->d_user, f);
       }
       if (hipMemcpy(ibuf.data(), (int64_t*)ix->d_ids.p + s0, c * 8,
                     hipMemcpyDeviceToHost) != hipSuccess ||
@@ -1871,12 +1906,14 @@ extern "C" dg_status dg_load(dg_index** out, const char* path,
     }
     if (ntotal > 0) {
       const int32_t M = desc.pq_m;
+      const int32_t dp = ix->desc.d;  // padded internal stride
       if ((st = dbuf_reserve(ix->d_ids, (size_t)ntotal * 8, ix->stream,
                              false)) != DG_OK ||
           (st = dbuf_reserve(ix->d_assign, (size_t)ntotal * 4, ix->stream,
                              false)) != DG_OK)
         break;
-      if (!is_pq && (st = dbuf_reserve(ix->d_vectors, (size_t)ntotal * d * 4,
+      if (!is_pq && (st = dbuf_reserve(ix->d_vectors,
+                                       (size_t)ntotal * dp * 4,
                                        ix->stream, false)) != DG_OK)
         break;
       if (is_pq && (st = dbuf_reserve(ix->d_codes, (size_t)ntotal * M,
@@ -1903,10 +1940,11 @@ extern "C" dg_status dg_load(dg_index** out, const char* path,
             st = DG_EIO;
             break;
           }
-          if (hipMemcpy((float*)ix->d_vectors.p + (size_t)s0 * d,
-                        vbuf.data(), c * d * 4,
-                        hipMemcpyHostToDevice) != hipSuccess)
+          if (upload_rows_padded(ix, (float*)ix->d_vectors.p +
+                                         (size_t)s0 * ix->desc.d,
+                                 vbuf.data(), (int64_t)c, false) != DG_OK)
             st = DG_EINTERNAL;
+          (void)hipStreamSynchronize(ix->stream);
         }
         if (fread(ibuf.data(), 8, c, f) != c ||
             fread(abuf.data(), 4, c, f) != c) {
@@ -1956,7 +1994,7 @@ extern "C" dg_status dg_stats(dg_index* ix, dg_stats_out* out) {
   DeviceGuard g(ix->device);
   memset(out, 0, sizeof(*out));
   out->ntotal = ix->ntotal - ix->n_deleted;
-  out->d = ix->desc.d;
+  out->d = ix->d_user;
   out->metric = ix->desc.metric;
   out->kind = ix->desc.kind;
   out->nlist = ix->desc.nlist;

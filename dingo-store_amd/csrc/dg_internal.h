@@ -76,7 +76,12 @@ struct dg_stage_times {
 };
 
 struct dg_index {
+  // INTERNAL dimension convention: desc.d is PADDED to a multiple of 4
+  // (zero-filled pad columns — dots and norms are unaffected); d_user is
+  // the caller's dimension, the row stride of every pointer crossing the
+  // ABI.  Boundary copies are 2D (pitched) when d_user != desc.d.
   dg_index_desc desc;
+  int32_t d_user = 0;
   int device = 0;
   hipStream_t stream = nullptr;
   rocblas_handle blas = nullptr;

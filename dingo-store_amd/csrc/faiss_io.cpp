@@ -153,7 +153,8 @@ struct HostRows {
 
 dg_status download_rows(dg_index* ix, HostRows& h) {
   const int64_t n = ix->ntotal;
-  const int32_t d = ix->desc.d;
+  const int32_t du = ix->d_user;   // file stride = caller's dimension
+  const int32_t dp = ix->desc.d;   // device stride (padded)
   const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
   h.ids.resize(n);
   h.assign.resize(n);
@@ -170,9 +171,10 @@ dg_status download_rows(dg_index* ix, HostRows& h) {
                   hipMemcpyDeviceToHost) != hipSuccess)
       return DG_EINTERNAL;
   } else {
-    h.vectors.resize((size_t)n * d);
-    if (hipMemcpy(h.vectors.data(), ix->d_vectors.p, h.vectors.size() * 4,
-                  hipMemcpyDeviceToHost) != hipSuccess)
+    h.vectors.resize((size_t)n * du);
+    if (hipMemcpy2D(h.vectors.data(), (size_t)du * 4, ix->d_vectors.p,
+                    (size_t)dp * 4, (size_t)du * 4, (size_t)n,
+                    hipMemcpyDeviceToHost) != hipSuccess)
       return DG_EINTERNAL;
   }
   return DG_OK;
@@ -285,7 +287,7 @@ extern "C" dg_status dg_save_faiss(dg_index* ix, const char* path) {
   if (!ix || !path) return DG_EINVAL;
   std::shared_lock lk(ix->rw);
   DeviceGuard g{ix->device};
-  const int32_t d = ix->desc.d;
+  const int32_t d = ix->d_user;  // containers speak the caller's dimension
   const int fm = to_faiss_metric(ix->desc.metric);
   const bool is_pq = ix->desc.kind == DG_INDEX_IVF_PQ;
   const bool is_flat = ix->desc.kind == DG_INDEX_FLAT;
@@ -324,8 +326,9 @@ extern "C" dg_status dg_save_faiss(dg_index* ix, const char* path) {
   } else {
     const int32_t nlist = ix->desc.nlist;
     std::vector<float> cents((size_t)nlist * d);
-    if (hipMemcpy(cents.data(), ix->d_centroids.p, cents.size() * 4,
-                  hipMemcpyDeviceToHost) != hipSuccess) {
+    if (hipMemcpy2D(cents.data(), (size_t)d * 4, ix->d_centroids.p,
+                    (size_t)ix->desc.d * 4, (size_t)d * 4, (size_t)nlist,
+                    hipMemcpyDeviceToHost) != hipSuccess) {
       fclose(f);
       return DG_EINTERNAL;
     }
