@@ -2045,14 +2045,24 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
       // but LLVM's allocator cannot be made to respect in-flight asm-load
       // destinations (verified by static hazard scan + GPU faults); the
       // drain form is airtight and measured fastest of the correct set.
-      constexpr int QTM = 16;
-      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-      hipLaunchKernelGGL((k_ivf_scan_pipe<QTM, true>),
-                         dim3((uint32_t)n_units), dim3(256), lds, s, units,
-                         csr_offsets, chunk_off, chunk_base, tvec, vnorms,
-                         queries, d, inv_offsets, inv_q, inv_rank, qp_off,
-                         q_cand_base, nprobe, metric, bitmap, chunk_rows,
-                         cand);
+      // QTM shrinks for large d so the query tile fits the 160 KB LDS.
+#define DG_PIPE_LAUNCH(QTM)                                                 \
+  do {                                                                      \
+    size_t lds = (size_t)(QTM) * d * 4 + (QTM) * 8;                         \
+    hipLaunchKernelGGL((k_ivf_scan_pipe<(QTM), true>),                      \
+                       dim3((uint32_t)n_units), dim3(256), lds, s, units,   \
+                       csr_offsets, chunk_off, chunk_base, tvec, vnorms,    \
+                       queries, d, inv_offsets, inv_q, inv_rank, qp_off,    \
+                       q_cand_base, nprobe, metric, bitmap, chunk_rows,     \
+                       cand);                                               \
+  } while (0)
+      if (d <= 2304)
+        DG_PIPE_LAUNCH(16);
+      else if (d <= 4608)
+        DG_PIPE_LAUNCH(8);
+      else
+        DG_PIPE_LAUNCH(4);  // d <= 8192 (create-time cap)
+#undef DG_PIPE_LAUNCH
       break;
     }
     case 15: {  // round-1 columnar pipeline (previous default)
