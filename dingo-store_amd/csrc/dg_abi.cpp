@@ -9,6 +9,7 @@
 #include <cstring>
 #include <algorithm>
 #include <functional>
+#include <atomic>
 #include <vector>
 
 #include "dg_internal.h"
@@ -40,12 +41,18 @@ extern "C" const char* dg_build_info(void) {
 }
 
 // ---------------- device buffer helpers ----------------
+// process-wide allocation generation: any (re)allocation may move a buffer
+// a captured hipGraph references, so graph caches key on this too
+// (over-invalidation is harmless — it just recaptures)
+static std::atomic<uint64_t> g_alloc_gen{1};
+
 static dg_status dbuf_reserve(dg_dbuf& b, size_t bytes, hipStream_t s,
                               bool keep) {
   if (bytes <= b.cap) {
     b.bytes = bytes;
     return DG_OK;
   }
+  g_alloc_gen.fetch_add(1, std::memory_order_relaxed);
   size_t newcap = std::max(bytes, b.cap + b.cap / 2);
   void* np = nullptr;
   if (hipMalloc(&np, newcap) != hipSuccess) {
@@ -328,8 +335,10 @@ extern "C" void dg_index_destroy(dg_index* ix) {
         &ix->d_codebooks, &ix->d_codes, &ix->d_csr_codes, &ix->d_S,
         &ix->d_cb_norms, &ix->ws_T, &ix->ws_Tf32, &ix->ws_queries, &ix->ws_qnorms,
         &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
-        &ix->ws_small, &ix->ws_topk, &ix->ws_scan, &ix->ws_seg})
+        &ix->ws_small, &ix->ws_topk, &ix->ws_scan, &ix->ws_seg, &ix->ws_gq,
+        &ix->ws_gout})
     dbuf_free(*b);
+  if (ix->graph_exec) (void)hipGraphExecDestroy(ix->graph_exec);
   for (auto& e : ix->ev)
     if (e) (void)hipEventDestroy(e);
   if (ix->h_pinned) (void)hipHostFree(ix->h_pinned);
@@ -367,6 +376,7 @@ extern "C" dg_status dg_set_centroids(dg_index* ix, int32_t nlist,
   DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
   ix->trained = true;
   ix->csr_valid = false;
+  ix->index_gen++;
   return DG_OK;
 }
 
@@ -583,6 +593,7 @@ extern "C" dg_status dg_set_codebooks(dg_index* ix, int32_t m, int32_t nbits,
   DG_HIP_CHECK(hipStreamSynchronize(ix->stream));
   ix->pq_trained = true;
   ix->csr_valid = false;
+  ix->index_gen++;
   return DG_OK;
 }
 
@@ -693,6 +704,7 @@ extern "C" dg_status dg_train(dg_index* ix, int64_t n, const float* x) {
   if (st == DG_OK) {
     ix->trained = true;
     ix->csr_valid = false;
+  ix->index_gen++;
   }
   return st;
 }
@@ -836,6 +848,7 @@ static dg_status add_impl(dg_index* ix, int64_t n, const int64_t* ids,
   for (int64_t i = 0; i < n; i++) ix->id_count.emplace(ids[i], 1);
   ix->ntotal += n;
   ix->csr_valid = false;
+  ix->index_gen++;
   return DG_OK;
 }
 
@@ -889,6 +902,7 @@ dg_status dg_ingest_rows(dg_index* ix, int64_t n, const int64_t* ids,
     if (ids[i] >= 0) ix->id_count.emplace(ids[i], 1);
   ix->ntotal += n;
   ix->csr_valid = false;
+  ix->index_gen++;
   return DG_OK;
 }
 
@@ -940,6 +954,7 @@ extern "C" dg_status dg_remove(dg_index* ix, int64_t n, const int64_t* ids) {
   for (int64_t i = 0; i < n; i++) ix->id_count.erase(ids[i]);
   ix->n_deleted += n;
   ix->csr_valid = false;
+  ix->index_gen++;
   return DG_OK;
 }
 
@@ -1123,6 +1138,7 @@ static dg_status finalize_csr(dg_index* ix) {
   }
   dbuf_free(rm_tmp);
   ix->csr_valid = true;
+  ix->index_gen++;
   return DG_OK;
 }
 
@@ -1152,7 +1168,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
   const int32_t nlist = is_ivf ? ix->desc.nlist : 1;
   dg_status st = DG_OK;
 
-  (void)hipEventRecord(ix->ev[0], ix->stream);
+  if (!ix->capturing) (void)hipEventRecord(ix->ev[0], ix->stream);
 
   // --- queries: copy (never mutate caller buffer), cosine-normalize, norms
   if ((st = dbuf_reserve(ix->ws_queries, (size_t)nq * d * 4, ix->stream,
@@ -1343,8 +1359,8 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     uint64_t* slab = (uint64_t*)ix->ws_topk.p;          // nq x slab_k
     uint64_t* final_tk = slab + (size_t)nq * slab_k;    // nq x k
     int64_t* bt = (int64_t*)(final_tk + (size_t)nq * k);   // base/total
-    (void)hipEventRecord(ix->ev[1], ix->stream);  // no coarse stage in Flat
-    (void)hipEventRecord(ix->ev[2], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[1], ix->stream);  // no coarse stage in Flat
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[2], ix->stream);
     const int mode = metric == DG_METRIC_L2 ? 1 : 2;
     if (rr) {
       // range search: count pass + compact pass (GEMM recomputed — range
@@ -1407,7 +1423,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                         nseg, k, mode, d_bitmap, c0, slab, slab_k,
                         ci * nseg * k);
     }
-    (void)hipEventRecord(ix->ev[3], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[3], ix->stream);
     if (st == DG_OK) {
       uint64_t* result = slab;
       if (slab_k > k) {
@@ -1419,7 +1435,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                         dqn, nq, k, metric, metric == DG_METRIC_L2 ? 1 : 0,
                         d_out_dist, d_out_ids);
     }
-    (void)hipEventRecord(ix->ev[4], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[4], ix->stream);
     ix->times.last_scan_bytes_alg = (int64_t)ix->ntotal * (d * 4 + 4);
   } else {
     // ---------- IVF ----------
@@ -1495,7 +1511,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
       }
       dgk::probe_unpack(ix->stream, coarse_tk, nq, np, maskp, probes);
     }
-    (void)hipEventRecord(ix->ev[1], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[1], ix->stream);
 
     // inverted mapping + candidate offsets
     size_t inv_bytes = (size_t)nlist * 4 * 3 + ((size_t)nlist + 1) * 12 +
@@ -1552,7 +1568,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     // algorithmic bytes (roofline) summed on device, read back async into
     // pinned memory; dg_stats synchronizes on ev[5]
     const int64_t row_bytes = is_pq ? ix->desc.pq_m : (int64_t)(d * 4 + 4);
-    if (ix->h_pinned) {
+    if (ix->h_pinned && !ix->capturing) {
       dg_dbuf& wsu = ix->ws_units;
       if ((st = dbuf_reserve(wsu, 8, ix->stream, false)) != DG_OK) {
         return st;
@@ -1563,10 +1579,10 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                      (int64_t*)wsu.p);
       (void)hipMemcpyAsync(ix->h_pinned, wsu.p, 8, hipMemcpyDeviceToHost,
                            ix->stream);
-      (void)hipEventRecord(ix->ev[5], ix->stream);
+      if (!ix->capturing) (void)hipEventRecord(ix->ev[5], ix->stream);
     }
 
-    (void)hipEventRecord(ix->ev[2], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[2], ix->stream);
     if (is_pq) {
       // ADC scan: build T (strided-batched GEMM, q_sub x codebook^T per
       // subspace) then gather-scan codes
@@ -1612,7 +1628,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
                         np, metric, d_bitmap, chunk_rows,
                         (uint64_t*)ix->ws_cand.p);
     }
-    (void)hipEventRecord(ix->ev[3], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[3], ix->stream);
 
     if (rr) {
       // range search over the candidate buffer
@@ -1644,7 +1660,7 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     dgk::emit_results(ix->stream, final_tk, (const int64_t*)ix->d_csr_ids.p,
                       dqn, nq, k, metric, metric == DG_METRIC_L2 ? 1 : 0,
                       d_out_dist, d_out_ids);
-    (void)hipEventRecord(ix->ev[4], ix->stream);
+    if (!ix->capturing) (void)hipEventRecord(ix->ev[4], ix->stream);
   }
   ix->times.last_nq = nq;
   return st;
@@ -1689,6 +1705,82 @@ extern "C" dg_status dg_search_device(dg_index* ix, int64_t nq,
   }
   std::shared_lock lk(ix->rw);
   std::lock_guard sg(ix->search_mu);  // workspaces + stream are shared
+
+  // ---- small-batch hipGraph path (nq=1 latency).
+  // The reference's pool issues nq=1 per Search call (vector_index.cc:53),
+  // so the fixed ~0.6 ms of launch overhead dominates that shape.  The
+  // kernel chain is captured once over fixed staging buffers and replayed;
+  // any mutation, finalize, or workspace reallocation bumps a generation
+  // and forces recapture.  Capture failure falls back to the normal path.
+  const bool graph_ok = nq <= 4 && k <= 128 &&
+                        (!filter || filter->kind == DG_FILTER_NONE) &&
+                        getenv("DG_NO_GRAPH") == nullptr;
+  if (graph_ok) {
+    const uint64_t gen =
+        ix->index_gen + g_alloc_gen.load(std::memory_order_relaxed);
+    const size_t q_bytes = (size_t)nq * ix->d_user * 4;
+    const size_t dist_bytes = (size_t)nq * k * 4;
+    const size_t ids_bytes = (size_t)nq * k * 8;
+    if (ix->graph_exec && ix->graph_nq == (int32_t)nq &&
+        ix->graph_k == k && ix->graph_np == nprobe &&
+        ix->graph_gen == gen) {
+      DG_HIP_CHECK(hipMemcpyAsync(ix->ws_gq.p, d_x, q_bytes,
+                                  hipMemcpyDeviceToDevice, ix->stream));
+      DG_HIP_CHECK(hipGraphLaunch(ix->graph_exec, ix->stream));
+      DG_HIP_CHECK(hipMemcpyAsync(d_out_dist, ix->ws_gout.p, dist_bytes,
+                                  hipMemcpyDeviceToDevice, ix->stream));
+      DG_HIP_CHECK(hipMemcpyAsync(
+          d_out_ids, (char*)ix->ws_gout.p + dist_bytes, ids_bytes,
+          hipMemcpyDeviceToDevice, ix->stream));
+      ix->times.last_nq = 0;  // per-stage timings not recorded on replays
+      return DG_OK;
+    }
+    // normal run first (sizes every workspace for this shape), then try to
+    // capture the same shape over the staging buffers
+    dg_status st = search_core(ix, nq, d_x, k, nprobe, filter, d_out_dist,
+                               d_out_ids);
+    if (st != DG_OK) return st;
+    if (dbuf_reserve(ix->ws_gq, q_bytes, ix->stream, false) != DG_OK ||
+        dbuf_reserve(ix->ws_gout, dist_bytes + ids_bytes, ix->stream,
+                     false) != DG_OK)
+      return DG_OK;  // no staging = no graph; result already produced
+    if (ix->graph_exec) {
+      (void)hipGraphExecDestroy(ix->graph_exec);
+      ix->graph_exec = nullptr;
+    }
+    (void)hipStreamSynchronize(ix->stream);
+    const uint64_t gen2 =
+        ix->index_gen + g_alloc_gen.load(std::memory_order_relaxed);
+    ix->capturing = true;
+    hipGraph_t graph = nullptr;
+    bool captured = false;
+    if (hipStreamBeginCapture(ix->stream, hipStreamCaptureModeThreadLocal) ==
+        hipSuccess) {
+      dg_status cst = search_core(
+          ix, nq, (const float*)ix->ws_gq.p, k, nprobe, nullptr,
+          (float*)ix->ws_gout.p,
+          (int64_t*)((char*)ix->ws_gout.p + dist_bytes));
+      if (hipStreamEndCapture(ix->stream, &graph) == hipSuccess &&
+          cst == DG_OK && graph) {
+        hipGraphExec_t ge = nullptr;
+        if (hipGraphInstantiate(&ge, graph, nullptr, nullptr, 0) ==
+            hipSuccess) {
+          ix->graph_exec = ge;
+          ix->graph_nq = (int32_t)nq;
+          ix->graph_k = k;
+          ix->graph_np = nprobe;
+          // capture itself allocates nothing (warm run sized buffers),
+          // so gen2 still describes the captured pointers
+          ix->graph_gen = gen2;
+          captured = true;
+        }
+      }
+      if (graph) (void)hipGraphDestroy(graph);
+    }
+    if (!captured) (void)hipGetLastError();  // clear capture errors
+    ix->capturing = false;
+    return DG_OK;
+  }
   return search_core(ix, nq, d_x, k, nprobe, filter, d_out_dist, d_out_ids);
 }
 
@@ -1962,6 +2054,7 @@ extern "C" dg_status dg_load(dg_index** out, const char* path,
       ix->ntotal = ntotal;
       ix->n_deleted = ndel;
       ix->csr_valid = false;
+  ix->index_gen++;
     }
   } while (0);
   fclose(f);

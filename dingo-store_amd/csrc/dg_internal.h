@@ -152,6 +152,17 @@ struct dg_index {
   dg_stage_times times;
   bool events_ready = false;
 
+  // ---- small-batch hipGraph cache (nq=1 latency path; the reference's
+  // thread pool issues nq=1 per Search call, vector_index.cc:53-54) ----
+  // Captured over search_core with fixed staging in/out buffers; replayed
+  // when (nq, k, nprobe) match and the index generation is unchanged.
+  hipGraphExec_t graph_exec = nullptr;
+  int32_t graph_nq = 0, graph_k = 0, graph_np = 0;
+  uint64_t graph_gen = 0;   // generation the graph was captured at
+  uint64_t index_gen = 1;   // bumped on finalize/mutation/buffer growth
+  dg_dbuf ws_gq, ws_gout;   // fixed staging: queries in, dist+ids out
+  bool capturing = false;   // search_core: skip event records in capture
+
   std::shared_mutex rw;  // search shared; mutation exclusive
   // Concurrent dg_search calls are SAFE but serialized per index: searches
   // share the workspace buffers and the HIP stream, so execution holds this

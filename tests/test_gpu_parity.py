@@ -245,6 +245,32 @@ def test_large_dimension_ivf():
     assert ids_match_with_tie_slack(gd, gi, od, oi) >= 0.98
 
 
+def test_small_batch_graph_replay():
+    # nq=1 path: call 2 captures a hipGraph, later calls replay it; results
+    # must stay oracle-identical, and mutation must invalidate the graph
+    base, q = make_data(n=20000, d=128, nq=8)
+    nlist, nprobe, k = 64, 16, 10
+    gpu, (cents, off, gv, gi_) = build_pair(orc.L2, base, nlist)
+    try:
+        for i in range(6):  # normal -> capture -> replays
+            gd, gi = gpu.search(q[i:i + 1], k, nprobe=nprobe)
+            od, oi = orc.ivf_search(orc.L2, cents, off, gv, gi_,
+                                    q[i:i + 1], k, nprobe)
+            assert ids_match_with_tie_slack(gd, gi, od, oi) >= 0.99, i
+        # mutation invalidates the captured graph
+        extra = base[:1] + 0.25
+        gpu.add(np.array([10_000_000], dtype=np.int64), extra)
+        gd, gi = gpu.search(extra, k, nprobe=nlist)
+        assert gi[0, 0] == 10_000_000  # self top-1 via full sweep
+        # and replays after re-capture still match the oracle
+        gd, gi = gpu.search(q[:1], k, nprobe=nprobe)
+        gd2, gi2 = gpu.search(q[:1], k, nprobe=nprobe)
+        np.testing.assert_array_equal(gi, gi2)
+        np.testing.assert_allclose(gd, gd2, rtol=0, atol=0)
+    finally:
+        gpu.close()
+
+
 def test_ivf_nprobe_default_and_clamp():
     # nprobe<=0 -> default 80 clamped to nlist (ivf_flat.cc:208-214,234)
     base, q = make_data(n=5000, d=64, nq=16)
