@@ -652,20 +652,31 @@ def test_concurrent_search():
 
 # ---------------- error paths (round-1 limits fail loudly) ----------------
 def test_error_paths():
-    with pytest.raises(dg.DgError):  # d % 4 != 0
-        dg.Index(dg.FLAT, dg.L2, 30)
-    with pytest.raises(dg.DgError):  # IVF d > 2048
-        dg.Index(dg.IVF_FLAT, dg.L2, 4096, nlist=16)
+    # round-2 limits: d <= 8192 (any d), k <= 2048, nprobe <= 2048 (< nlist)
+    with pytest.raises(dg.DgError):  # d > 8192
+        dg.Index(dg.FLAT, dg.L2, 8200)
     with pytest.raises(dg.DgError):  # PQ d % m != 0
         dg.Index(dg.IVF_PQ, dg.L2, 64, nlist=16, m=7)
+    with pytest.raises(dg.DgError):  # PQ d % 4 != 0 (padding excluded)
+        dg.Index(dg.IVF_PQ, dg.L2, 66, nlist=16, m=6)
     base, q = make_data(n=2000, d=32, nq=4)
     idx = dg.Index(dg.FLAT, dg.L2, 32)
     try:
         idx.add(np.arange(2000, dtype=np.int64), base)
-        with pytest.raises(dg.DgError):  # k > 128 unsupported this round
-            idx.search(q, 200)
+        with pytest.raises(dg.DgError):  # k > 2048 unsupported
+            idx.search(q, 2100)
     finally:
         idx.close()
+    big = dg.Index(dg.IVF_FLAT, dg.L2, 32, nlist=4096)
+    try:
+        with pytest.raises(dg.DgError):  # nprobe > 2048 and < nlist
+            cents = np.zeros((4096, 32), np.float32)
+            cents[:, 0] = np.arange(4096)
+            big.set_centroids(cents)
+            big.add(np.arange(100, dtype=np.int64), base[:100])
+            big.search(q, 5, nprobe=3000)
+    finally:
+        big.close()
     gpu, _ = build_pair(orc.L2, base, 16)
     try:
         gd, gi = gpu.search(q, 5, nprobe=0)  # default nprobe path
