@@ -1620,13 +1620,15 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
       const int32_t* d_chunk_off = (const int32_t*)mp;
       const int64_t* d_chunk_base =
           (const int64_t*)(mp + ((size_t)nlist + 1) * 4);
+      const int32_t mean_probes =
+          (int32_t)((nq * (int64_t)np) / std::max(1, nlist));
       dgk::ivf_scan_col(ix->stream, units, total_units,
                         (const int64_t*)ix->d_csr_offsets.p, d_chunk_off,
                         d_chunk_base, (const float*)ix->d_csr_t.p,
                         (const float*)ix->d_csr_vnorms.p, dq, d,
                         inv_offsets32, inv_q, inv_rank, qp_off, q_cand_base,
                         np, metric, d_bitmap, chunk_rows,
-                        (uint64_t*)ix->ws_cand.p);
+                        (uint64_t*)ix->ws_cand.p, mean_probes);
     }
     if (!ix->capturing) (void)hipEventRecord(ix->ev[3], ix->stream);
 

@@ -232,7 +232,8 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                   const int32_t* inv_offsets, const int32_t* inv_q,
                   const int32_t* inv_rank, const int64_t* qp_off,
                   const int64_t* q_cand_base, int32_t nprobe, int metric,
-                  const uint32_t* bitmap, int32_t chunk_rows, uint64_t* cand);
+                  const uint32_t* bitmap, int32_t chunk_rows, uint64_t* cand,
+                  int32_t mean_probes);
 void transpose_chunks(hipStream_t s, const uint32_t* units, int32_t n_units,
                       const int64_t* csr_offsets, const int32_t* chunk_off,
                       const int64_t* chunk_base, const float* rowmajor,

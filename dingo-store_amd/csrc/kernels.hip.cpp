@@ -1954,7 +1954,7 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                   const int32_t* inv_rank, const int64_t* qp_off,
                   const int64_t* q_cand_base, int32_t nprobe, int metric,
                   const uint32_t* bitmap, int32_t chunk_rows,
-                  uint64_t* cand) {
+                  uint64_t* cand, int32_t mean_probes) {
   if (!n_units) return;
   static int variant = []() {
     const char* e = getenv("DG_SCAN_VARIANT");
@@ -2056,7 +2056,9 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                        q_cand_base, nprobe, metric, bitmap, chunk_rows,     \
                        cand);                                               \
   } while (0)
-      if (d <= 2304)
+      if (d <= 1530 && mean_probes > 16)
+        DG_PIPE_LAUNCH(24);  // dense batches: fewer chunk re-streams
+      else if (d <= 2304)
         DG_PIPE_LAUNCH(16);
       else if (d <= 4608)
         DG_PIPE_LAUNCH(8);
