@@ -1107,6 +1107,164 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
   }
 }
 
+// ---------- generalized drain pipeline (v5, DG_SCAN_VARIANT=16..18) ----
+// SAFE-mode generalization of k_ivf_scan_pipe: S stages x 4 dims issued as
+// one batch at block end, ONE vmcnt(0) drain at the next block's head
+// (tied per stage before its compute), amortizing the exposed HBM latency
+// over 4*S dims instead of 4.  All waits are full drains, so the LLVM
+// in-flight-register hazards of the counted variants cannot occur.
+template <int QTM, int S>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe3(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  constexpr int RPL = 4;
+  extern __shared__ __attribute__((aligned(16))) float smem[];  // [QTM * d]
+  int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);          // [QTM]
+
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* col = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+  const int64_t row0 = list_start + (int64_t)chunk * chunk_rows;
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  __builtin_assume(d % 4 == 0 && d > 0);
+
+  for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
+    const int32_t qt = min(QTM, nql - t0);
+    __syncthreads();
+    for (int32_t j = 0; j < qt; j++) {
+      int32_t q = inv_q[iq0 + t0 + j];
+      const float4* src = (const float4*)(queries + (size_t)q * d);
+      float4* dst = (float4*)(smem + (size_t)j * d);
+      for (int i = threadIdx.x; i < d / 4; i += blockDim.x) dst[i] = src[i];
+    }
+    for (size_t i = (size_t)qt * d + threadIdx.x; i < (size_t)QTM * d;
+         i += blockDim.x)
+      smem[i] = 0.f;
+    if (threadIdx.x < QTM) {
+      int32_t j = threadIdx.x;
+      if (j < qt) {
+        int32_t q = inv_q[iq0 + t0 + j];
+        int32_t rank = inv_rank[iq0 + t0 + j];
+        cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                   list_start;
+      } else {
+        cbase[j] = 0;
+      }
+    }
+    __syncthreads();
+
+    for (int32_t rb = wave_id * WAVE * RPL; rb < nrows_pad;
+         rb += 4 * WAVE * RPL) {
+      const int32_t rr0 = rb + lane * RPL;
+      if (rr0 >= nrows_pad) continue;
+      float acc[QTM][RPL];
+#pragma unroll
+      for (int j = 0; j < QTM; j++)
+#pragma unroll
+        for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
+
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // prior emit stores
+      const char* colp0 = (const char*)(col + rr0);
+      const size_t cstride = (size_t)nrows_pad * 4;
+      dg_f4 buf[S][4];
+      auto issue_block = [&](int32_t ib) {
+#pragma unroll
+        for (int i = 0; i < S; i++)
+#pragma unroll
+          for (int u = 0; u < 4; u++)
+            DG_GLOAD4(buf[i][u],
+                      colp0 + (size_t)(ib + 4 * i + u) * cstride);
+      };
+      auto compute4 = [&](const dg_f4 (&b)[4], int32_t ib) {
+#pragma unroll
+        for (int j = 0; j < QTM; j++) {
+          const float* qp = smem + (size_t)j * d + ib;
+          const float2 qa = *(const float2*)qp;
+          const float2 qb = *(const float2*)(qp + 2);
+          const float qv[4] = {qa.x, qa.y, qb.x, qb.y};
+#pragma unroll
+          for (int u = 0; u < 4; u++)
+#pragma unroll
+            for (int x = 0; x < RPL; x++) acc[j][x] += b[u][x] * qv[u];
+        }
+      };
+      constexpr int BD = 4 * S;  // dims per block
+      int32_t base = 0;
+      if (d >= BD) {
+        issue_block(0);
+        for (; base + BD <= d; base += BD) {
+          const bool more = base + 2 * BD <= d;
+#pragma unroll
+          for (int i = 0; i < S; i++) {
+            DG_WAITV(0, buf[i]);  // first drains everything; rest order
+            compute4(buf[i], base + 4 * i);
+          }
+          if (more) issue_block(base + BD);
+        }
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      // tail < BD dims: compiler-managed loads
+      const float* colf = col + rr0;
+      for (int32_t ib = base; ib < d; ib += 4) {
+        float4 c[4];
+#pragma unroll
+        for (int u = 0; u < 4; u++)
+          c[u] = (ib + u < d)
+                     ? *(const float4*)(colf + (size_t)(ib + u) * nrows_pad)
+                     : float4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int j = 0; j < QTM; j++) {
+          const float* qp = smem + (size_t)j * d + ib;
+#pragma unroll
+          for (int u = 0; u < 4 && ib + u < d; u++) {
+            const float qv = qp[u];
+            acc[j][0] += c[u].x * qv;
+            acc[j][1] += c[u].y * qv;
+            acc[j][2] += c[u].z * qv;
+            acc[j][3] += c[u].w * qv;
+          }
+        }
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+#pragma unroll
+      for (int j = 0; j < QTM; j++) {
+        if (j < qt) {
+          const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+          for (int x = 0; x < RPL; x++) {
+            const int32_t rl = rr0 + x;
+            if (rl < nrows) {
+              const int64_t r = row0 + rl;
+              bool pass = true;
+              if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+              float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                        : -acc[j][x];
+              cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
 // ---------- fused-stage asm scan (v4, DG_SCAN_VARIANT=13) ----------
 // The v3 kernel's separate issue/wait asms are not airtight: LLVM may copy
 // or reuse an in-flight load's destination registers between the asm
@@ -2009,36 +2167,6 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                          cand);
       break;
     }
-    case 10: {  // asm-pipelined counted-vmcnt scan (v3)
-      constexpr int QTM = 16;
-      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-      hipLaunchKernelGGL((k_ivf_scan_pipe<QTM>), dim3((uint32_t)n_units),
-                         dim3(256), lds, s, units, csr_offsets, chunk_off,
-                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
-                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
-                         metric, bitmap, chunk_rows, cand);
-      break;
-    }
-    case 13: {  // fused-stage asm scan (v4)
-      constexpr int QTM = 16;
-      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-      hipLaunchKernelGGL((k_ivf_scan_pipe2<QTM>), dim3((uint32_t)n_units),
-                         dim3(256), lds, s, units, csr_offsets, chunk_off,
-                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
-                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
-                         metric, bitmap, chunk_rows, cand);
-      break;
-    }
-    case 14: {  // fused-stage asm scan, QTM=12 (lower register pressure)
-      constexpr int QTM = 12;
-      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-      hipLaunchKernelGGL((k_ivf_scan_pipe2<QTM>), dim3((uint32_t)n_units),
-                         dim3(256), lds, s, units, csr_offsets, chunk_off,
-                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
-                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
-                         metric, bitmap, chunk_rows, cand);
-      break;
-    }
     case 0:
     case 12: {  // DEFAULT (round 2): asm-load pipeline with full-drain
       // waits — the counted-wait variants (10/13/14) are faster on paper
@@ -2071,16 +2199,9 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
       DG_SCAN_LAUNCH(16, 4);
       break;
     }
-    case 11: {  // asm-pipelined, QTM=12
-      constexpr int QTM = 12;
-      size_t lds = (size_t)QTM * d * 4 + QTM * 8;
-      hipLaunchKernelGGL((k_ivf_scan_pipe<QTM>), dim3((uint32_t)n_units),
-                         dim3(256), lds, s, units, csr_offsets, chunk_off,
-                         chunk_base, tvec, vnorms, queries, d, inv_offsets,
-                         inv_q, inv_rank, qp_off, q_cand_base, nprobe,
-                         metric, bitmap, chunk_rows, cand);
-      break;
-    }
+// (pipe2/pipe3 launch cases removed: every counted-wait or batched-issue
+// schedule fails the static in-flight-register hazard scan — kernels kept
+// above as the documented record of the attempt)
     case 9: {  // (12,4) capped to 3 waves/SIMD (uncapped allocates 177)
       constexpr int QTM = 12;
       size_t lds = (size_t)QTM * d * 4 + QTM * 8;
