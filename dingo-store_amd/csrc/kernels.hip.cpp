@@ -954,6 +954,108 @@ typedef float dg_f4 __attribute__((ext_vector_type(4)));
       DG_WAITV(n, b);                                               \
   } while (0)
 
+// row-scan body of k_ivf_scan_pipe, templated on the EFFECTIVE query-tile
+// width JT: tiles that fill at most half the QTM slots take the JT=QTM/2
+// instantiation and skip the zero-padded FMAs (at cfg C the mean tile
+// holds ~8 of 16 queries — half the compute multiplies zeros otherwise).
+template <int JT, bool SAFE>
+__device__ __forceinline__ void dg_scan_rows_body(
+    const float* smem, const int64_t* cbase, const float* col,
+    const float* __restrict__ vnorms, int32_t d, int32_t nrows,
+    int32_t nrows_pad, int32_t rr0, int64_t row0, int32_t qt, int metric,
+    const uint32_t* __restrict__ bitmap, uint64_t* __restrict__ cand) {
+  constexpr int RPL = 4;
+  constexpr int U = 4;
+  float acc[JT][RPL];
+#pragma unroll
+  for (int j = 0; j < JT; j++)
+#pragma unroll
+    for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
+
+  // drain prior vmem (previous tile's emit stores) so vmcnt counts below
+  // track ONLY this loop's column loads
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const char* colp = (const char*)(col + rr0);
+  const size_t cstride = (size_t)nrows_pad * 4;  // bytes per dim column
+  dg_f4 b0[U], b1[U], b2[U], b3[U];
+  auto issue = [&](dg_f4 (&b)[U], int32_t ib) {
+    const char* p = colp + (size_t)ib * cstride;
+#pragma unroll
+    for (int u = 0; u < U; u++) DG_GLOAD4(b[u], p + u * cstride);
+  };
+  auto compute = [&](dg_f4 (&b)[U], int32_t ib) {
+#pragma unroll
+    for (int j = 0; j < JT; j++) {
+      const float* qp = smem + (size_t)j * d + ib;
+      const float2 qa = *(const float2*)qp;
+      const float2 qb = *(const float2*)(qp + 2);
+      const float qv[4] = {qa.x, qa.y, qb.x, qb.y};
+#pragma unroll
+      for (int u = 0; u < U; u++)
+#pragma unroll
+        for (int x = 0; x < RPL; x++) acc[j][x] += b[u][x] * qv[u];
+    }
+  };
+  // modulo schedule, period 4 stages = 16 dims
+  issue(b0, 0);
+  issue(b1, U);
+  issue(b2, 2 * U);
+  issue(b3, 3 * U);
+  int32_t base = 0;
+  for (; base + 16 < d; base += 16) {
+    DG_WAITV_N(12, b0);
+    compute(b0, base);
+    issue(b0, base + 16);
+    DG_WAITV_N(12, b1);
+    compute(b1, base + 4);
+    issue(b1, base + 20);
+    DG_WAITV_N(12, b2);
+    compute(b2, base + 8);
+    issue(b2, base + 24);
+    DG_WAITV_N(12, b3);
+    compute(b3, base + 12);
+    issue(b3, base + 28);
+  }
+  {  // epilogue: r = d - base in {4, 8, 12, 16} (d % 4 == 0)
+    const int32_t r = d - base;
+    DG_WAITV_N(12, b0);
+    compute(b0, base);
+    if (r > 4) {
+      DG_WAITV_N(8, b1);
+      compute(b1, base + 4);
+    }
+    if (r > 8) {
+      DG_WAITV_N(4, b2);
+      compute(b2, base + 8);
+    }
+    if (r > 12) {
+      DG_WAITV(0, b3);
+      compute(b3, base + 12);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+#pragma unroll
+  for (int j = 0; j < JT; j++) {
+    if (j < qt) {
+      const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+      for (int x = 0; x < RPL; x++) {
+        const int32_t rl = rr0 + x;
+        if (rl < nrows) {
+          const int64_t r = row0 + rl;
+          bool pass = true;
+          if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+          float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                    : -acc[j][x];
+          cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+        }
+      }
+    }
+  }
+}
+
 template <int QTM, bool SAFE = false>
 __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
@@ -966,7 +1068,6 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
     const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
     uint64_t* __restrict__ cand) {
   constexpr int RPL = 4;  // rows per lane (float4 column loads)
-  constexpr int U = 4;    // dims per stage
   extern __shared__ __attribute__((aligned(16))) float smem[];  // [QTM * d]
   int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);          // [QTM]
 
@@ -1015,94 +1116,14 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
          rb += 4 * WAVE * RPL) {
       const int32_t rr0 = rb + lane * RPL;
       if (rr0 >= nrows_pad) continue;
-      float acc[QTM][RPL];
-#pragma unroll
-      for (int j = 0; j < QTM; j++)
-#pragma unroll
-        for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
-
-      // drain prior vmem (previous tile's emit stores) so vmcnt counts
-      // below track ONLY this loop's column loads
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-
-      const char* colp = (const char*)(col + rr0);
-      const size_t cstride = (size_t)nrows_pad * 4;  // bytes per dim column
-      dg_f4 b0[U], b1[U], b2[U], b3[U];
-      auto issue = [&](dg_f4 (&b)[U], int32_t ib) {
-        const char* p = colp + (size_t)ib * cstride;
-#pragma unroll
-        for (int u = 0; u < U; u++) DG_GLOAD4(b[u], p + u * cstride);
-      };
-      auto compute = [&](dg_f4 (&b)[U], int32_t ib) {
-#pragma unroll
-        for (int j = 0; j < QTM; j++) {
-          const float* qp = smem + (size_t)j * d + ib;
-          const float2 qa = *(const float2*)qp;
-          const float2 qb = *(const float2*)(qp + 2);
-          const float qv[4] = {qa.x, qa.y, qb.x, qb.y};
-#pragma unroll
-          for (int u = 0; u < U; u++)
-#pragma unroll
-            for (int x = 0; x < RPL; x++) acc[j][x] += b[u][x] * qv[u];
-        }
-      };
-      // modulo schedule, period 4 stages = 16 dims
-      issue(b0, 0);
-      issue(b1, U);
-      issue(b2, 2 * U);
-      issue(b3, 3 * U);
-      int32_t base = 0;
-      for (; base + 16 < d; base += 16) {
-        DG_WAITV_N(12, b0);
-        compute(b0, base);
-        issue(b0, base + 16);
-        DG_WAITV_N(12, b1);
-        compute(b1, base + 4);
-        issue(b1, base + 20);
-        DG_WAITV_N(12, b2);
-        compute(b2, base + 8);
-        issue(b2, base + 24);
-        DG_WAITV_N(12, b3);
-        compute(b3, base + 12);
-        issue(b3, base + 28);
-      }
-      {  // epilogue: r = d - base in {4, 8, 12, 16} (d % 4 == 0)
-        const int32_t r = d - base;
-        DG_WAITV_N(12, b0);
-        compute(b0, base);
-        if (r > 4) {
-          DG_WAITV_N(8, b1);
-          compute(b1, base + 4);
-        }
-        if (r > 8) {
-          DG_WAITV_N(4, b2);
-          compute(b2, base + 8);
-        }
-        if (r > 12) {
-          DG_WAITV(0, b3);
-          compute(b3, base + 12);
-        }
-      }
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-
-#pragma unroll
-      for (int j = 0; j < QTM; j++) {
-        if (j < qt) {
-          const int64_t cb = cbase[j] + row0 + rr0;
-#pragma unroll
-          for (int x = 0; x < RPL; x++) {
-            const int32_t rl = rr0 + x;
-            if (rl < nrows) {
-              const int64_t r = row0 + rl;
-              bool pass = true;
-              if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
-              float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
-                                        : -acc[j][x];
-              cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
-            }
-          }
-        }
-      }
+      if (QTM >= 16 && qt <= QTM / 2)
+        dg_scan_rows_body<(QTM >= 16 ? QTM / 2 : QTM), SAFE>(
+            smem, cbase, col, vnorms, d, nrows, nrows_pad, rr0, row0, qt,
+            metric, bitmap, cand);
+      else
+        dg_scan_rows_body<QTM, SAFE>(smem, cbase, col, vnorms, d, nrows,
+                                     nrows_pad, rr0, row0, qt, metric,
+                                     bitmap, cand);
     }
   }
 }
