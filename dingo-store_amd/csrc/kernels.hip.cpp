@@ -2205,9 +2205,11 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
                        q_cand_base, nprobe, metric, bitmap, chunk_rows,     \
                        cand);                                               \
   } while (0)
-      if (d <= 1530 && mean_probes > 16)
-        DG_PIPE_LAUNCH(24);  // dense batches: fewer chunk re-streams
-      else if (d <= 2304)
+      (void)mean_probes;  // dense-batch 20/24-query tiles fail the
+                          // in-flight hazard scan (allocator stashes
+                          // pending asm loads at >=250 VGPRs) — QTM=16
+                          // is the largest clean tile
+      if (d <= 2304)
         DG_PIPE_LAUNCH(16);
       else if (d <= 4608)
         DG_PIPE_LAUNCH(8);
