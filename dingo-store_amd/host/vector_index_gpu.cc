@@ -7,6 +7,7 @@
 #include "vector_index_gpu.h"
 
 #include <cmath>
+#include <cstdio>
 #include <cstring>
 
 #include <algorithm>
@@ -380,6 +381,8 @@ std::unique_ptr<VectorIndex> NewIvfPqIndex(MetricType metric, int32_t dim,
 // ---------------- self-test (GPU) ----------------
 extern "C" int dg_mirror_selftest(void) {
   using namespace dingogpu;
+#define DG_ST(x) fprintf(stderr, "[selftest] %s\n", x)
+  DG_ST("flat loop");
   const int32_t d = 64, n = 500;
   for (MetricType m :
        {MetricType::kL2, MetricType::kInnerProduct, MetricType::kCosine}) {
@@ -423,6 +426,7 @@ extern "C" int dg_mirror_selftest(void) {
       if (vd.vector_with_id.id >= 100) return 10;
   }
 
+  DG_ST("ivf lifecycle");
   // ---- IVF lifecycle through the mirror (Train/Add/Search/RangeSearch/
   // Upsert/Delete/Save/Load), L2 ----
   {
@@ -459,6 +463,7 @@ extern "C" int dg_mirror_selftest(void) {
     if (r1[0].vector_with_distances.empty() ||
         r1[0].vector_with_distances[0].vector_with_id.id != 7)
       return 30;
+    DG_ST("range search");
     // range search around the self-distance
     std::vector<VectorWithDistanceResult> r2;
     if (!ivf->RangeSearch({batch[7]}, 0.5f, {}, false, p, r2).ok())
@@ -467,6 +472,7 @@ extern "C" int dg_mirror_selftest(void) {
     for (auto& vd : r2[0].vector_with_distances)
       if (vd.vector_with_id.id == 7) has_self = true;
     if (!has_self) return 32;
+    DG_ST("delete+upsert");
     // delete + upsert
     if (!ivf->Delete({7}).ok()) return 33;
     int64_t delc = -1;
@@ -482,6 +488,7 @@ extern "C" int dg_mirror_selftest(void) {
     if (r4[0].vector_with_distances.empty() ||
         r4[0].vector_with_distances[0].vector_with_id.id != 7)
       return 36;
+    DG_ST("save/load");
     // save / load round trip
     const char* path = "/tmp/dg_mirror_selftest.dgi";
     if (!ivf->Save(path).ok()) return 37;
@@ -491,6 +498,7 @@ extern "C" int dg_mirror_selftest(void) {
     if (r5[0].vector_with_distances.empty() ||
         r5[0].vector_with_distances[0].vector_with_id.id != 7)
       return 39;
+    DG_ST("dim mismatch");
     // dimension mismatch rejected (CheckVectorDimension semantics)
     VectorWithId bad;
     bad.id = 999999;
@@ -498,6 +506,7 @@ extern "C" int dg_mirror_selftest(void) {
     bad.vector.float_values.resize(d2 / 2);
     if (ivf->Add({bad}).ok()) return 40;
 
+    DG_ST("lifecycle virtuals");
     // ---- wrapper-lifecycle virtuals (§8b must-implement set) ----
     if (!ivf->SupportSave()) return 41;
     if (ivf->IsExceedsMaxElements(1 << 20)) return 42;  // always false
@@ -508,6 +517,7 @@ extern "C" int dg_mirror_selftest(void) {
     if (ivf->NeedToRebuild()) return 46;      // ntotal < 256 * nlist
     ivf->LockWrite();   // exclusive fork-save window lock
     ivf->UnlockWrite();
+    DG_ST("raw-float train");
     // raw-float Train(std::vector<float>&) on a fresh index
     {
       auto ivf2 = NewIvfFlatIndex(MetricType::kL2, d2, 4);
@@ -522,6 +532,7 @@ extern "C" int dg_mirror_selftest(void) {
       if (!ivf2->Train(tdata).ok()) return 49;
       if (!ivf2->IsTrained()) return 50;
     }
+    DG_ST("concrete filter");
     // ConcreteFilterFunctor (IDSelectorBatch semantics) incl. Build hook
     {
       std::vector<int64_t> want{3, 9, 15};
@@ -548,5 +559,6 @@ extern "C" int dg_mirror_selftest(void) {
       }
     }
   }
+  DG_ST("done");
   return 0;
 }
