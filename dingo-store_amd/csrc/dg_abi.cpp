@@ -227,10 +227,16 @@ static dg_status sgemm_dots(dg_index* ix, const float* X, int64_t rows,
   }();
   // hand kernel for the coarse-assign class (measured 57 TF vs rocBLAS
   // 63 TF on the Flat 1M-col shape, so the library keeps the huge-col
-  // scans; DESIGN.md records the gap and the round-2 pipelining plan)
+  // scans; DESIGN.md records the gap).  Tiny row counts (nq <= 8, the
+  // latency path) also go to the hand kernel so hipGraph captures never
+  // contain rocBLAS: its internal workspace can be reallocated by a
+  // different-shaped GEMM between capture and replay, leaving the graph
+  // with stale pointers (measured: heap corruption in the mirror
+  // selftest's mutate-then-recapture sequence).
   const bool use_mfma =
       force != 1 &&
-      (force == 2 || (rows >= 48 && cols >= 64 && cols <= 16384 && d >= 64));
+      (force == 2 || rows <= 8 ||
+       (rows >= 48 && cols >= 64 && cols <= 16384 && d >= 64));
   if (use_mfma) {
     dgk::dots_mfma(ix->stream, X, rows, Y, cols, d, dots, cols);
     return DG_OK;
@@ -1714,7 +1720,12 @@ extern "C" dg_status dg_search_device(dg_index* ix, int64_t nq,
   // kernel chain is captured once over fixed staging buffers and replayed;
   // any mutation, finalize, or workspace reallocation bumps a generation
   // and forces recapture.  Capture failure falls back to the normal path.
+  // IVF-Flat only: its nq<=4 search contains no rocBLAS call (the tiny
+  // coarse GEMM routes to the hand MFMA kernel) — Flat's chunked dots and
+  // PQ's T-build are library GEMMs, which are not capture-safe (see
+  // sgemm_dots).
   const bool graph_ok = nq <= 4 && k <= 128 &&
+                        ix->desc.kind == DG_INDEX_IVF_FLAT &&
                         (!filter || filter->kind == DG_FILTER_NONE) &&
                         getenv("DG_NO_GRAPH") == nullptr;
   if (graph_ok) {
