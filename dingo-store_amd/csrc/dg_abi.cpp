@@ -342,7 +342,7 @@ extern "C" void dg_index_destroy(dg_index* ix) {
         &ix->d_cb_norms, &ix->ws_T, &ix->ws_Tf32, &ix->ws_queries, &ix->ws_qnorms,
         &ix->ws_dots, &ix->ws_probes, &ix->ws_inv, &ix->ws_cand, &ix->ws_units,
         &ix->ws_small, &ix->ws_topk, &ix->ws_scan, &ix->ws_seg, &ix->ws_gq,
-        &ix->ws_gout})
+        &ix->ws_gout, &ix->ws_bm})
     dbuf_free(*b);
   if (ix->graph_exec) (void)hipGraphExecDestroy(ix->graph_exec);
   for (auto& e : ix->ev)
@@ -1221,16 +1221,20 @@ static dg_status search_core(dg_index* ix, int64_t nq, const float* d_x,
     }
   }
   uint32_t* d_bitmap = nullptr;
-  dg_dbuf ws_bitmap{};
-  DbufGuard g_bitmap(ws_bitmap);
+  // ws_bm is INDEX-OWNED, not a per-search local: a hipGraph captured over
+  // a tombstone-filtering search records kernels that read this buffer, so
+  // it must outlive the capture (a local RAII buffer here was freed at
+  // capture end and replays wrote freed memory — measured heap corruption
+  // in the mirror selftest's delete-then-search sequence)
   if (need_bitmap && ix->ntotal > 0) {
     size_t words = (size_t)((ix->ntotal + 31) / 32);
-    if ((st = dbuf_reserve(ws_bitmap, words * 4, ix->stream, false)) != DG_OK) {
+    if ((st = dbuf_reserve(ix->ws_bm, words * 4, ix->stream, false)) !=
+        DG_OK) {
       return st;
     }
     dgk::build_pass_bitmap(ix->stream, (const int64_t*)ix->d_csr_ids.p,
-                           ix->ntotal, &df, (uint32_t*)ws_bitmap.p);
-    d_bitmap = (uint32_t*)ws_bitmap.p;
+                           ix->ntotal, &df, (uint32_t*)ix->ws_bm.p);
+    d_bitmap = (uint32_t*)ix->ws_bm.p;
   }
 
   // ---- range-search shared pieces ----

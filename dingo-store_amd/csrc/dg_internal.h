@@ -146,6 +146,9 @@ struct dg_index {
   dg_dbuf ws_scan;
   // segmented-select slab (wide-k coarse top-nprobe merge)
   dg_dbuf ws_seg;
+  // per-row pass bitmap (filters/tombstones) — index-owned so captured
+  // graphs may reference it (see search_core)
+  dg_dbuf ws_bm;
 
   // timing
   hipEvent_t ev[12] = {};

@@ -788,7 +788,10 @@ __global__ void k_f32_to_f16(const float* __restrict__ in, int64_t n,
 // vector and walks m = 0..M-1 sequentially, so all lanes of a wave gather
 // within the same 1 KB rows of T[q] and S[l] (L1-resident after first
 // touch).  No cross-lane reduction.
-template <int RPV, int MINB = 1>  // rows (vectors) per lane; TILE = RPV * 64
+// COOP: all 4 waves walk the SAME query over disjoint quarters of the
+// staged tile (one coherent 48 KB T-window in L1 at a time) instead of 4
+// different queries (4 windows thrash the 32 KB L1).
+template <int RPV, int MINB = 1, bool COOP = false>
 __global__ void __launch_bounds__(256, MINB) k_ivfpq_scan(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
     const uint8_t* __restrict__ csr_codes, const __half* __restrict__ S,
@@ -838,32 +841,35 @@ __global__ void __launch_bounds__(256, MINB) k_ivfpq_scan(
     }
     __syncthreads();
 
-    // waves split the probing queries; each LANE owns 4 of the 256 staged
-    // vectors (v, v+64, v+128, v+192), so every (wave, query, m) visit of a
-    // 1 KB table row is consumed by all 256 vectors — without this, 3 KB of
-    // S/T row traffic per (vector, query) pair is the bottleneck (S at
-    // cfg D is 1.6 GB, beyond L3).
-    for (int32_t qi = wave_id; qi < nql; qi += blockDim.x / WAVE) {
+    // waves split the probing queries; each LANE owns RPV of the staged
+    // vectors, so every (wave, query, m) visit of a 1 KB table row is
+    // consumed by the whole tile — without this, 3 KB of S/T row traffic
+    // per (vector, query) pair is the bottleneck (S at cfg D is 1.6 GB,
+    // beyond L3).  COOP instead walks queries jointly (see above).
+    constexpr int RV = COOP ? RPV / 4 : RPV;
+    const int32_t vbase = COOP ? wave_id * (TILE / 4) : 0;
+    for (int32_t qi = COOP ? 0 : wave_id; qi < nql;
+         qi += COOP ? 1 : blockDim.x / WAVE) {
       const int32_t q = inv_q[iq0 + qi];
       const int32_t rank = inv_rank[iq0 + qi];
       const __half* Tq = T + (size_t)q * M * 256;
       const float dot = coarse_dots[(size_t)q * nlist + list];
       const int64_t cb0 = q_cand_base[q] +
                           qp_off[(int64_t)q * nprobe + rank] - list_start;
-      float acc[RPV];
-      const uint32_t* code4[RPV];
+      float acc[RV];
+      const uint32_t* code4[RV];
 #pragma unroll
-      for (int r = 0; r < RPV; r++) {
+      for (int r = 0; r < RV; r++) {
         acc[r] = 0.f;
-        code4[r] =
-            (const uint32_t*)(lds_codes + (size_t)(lane + r * WAVE) * M);
+        code4[r] = (const uint32_t*)(lds_codes +
+                                     (size_t)(vbase + lane + r * WAVE) * M);
       }
       if (metric == 0) {
         const __half* Sm = Sl;
         const __half* Tm = Tq;
         for (int32_t m4 = 0; m4 < M / 4; m4++) {
 #pragma unroll
-          for (int r = 0; r < RPV; r++) {
+          for (int r = 0; r < RV; r++) {
             const uint32_t cw = code4[r][m4];
             const uint32_t c0 = cw & 255, c1 = (cw >> 8) & 255,
                            c2 = (cw >> 16) & 255, c3 = cw >> 24;
@@ -879,12 +885,12 @@ __global__ void __launch_bounds__(256, MINB) k_ivfpq_scan(
           Tm += 1024;
         }
 #pragma unroll
-        for (int r = 0; r < RPV; r++) acc[r] -= 2.0f * dot;  // +qnorm@emit
+        for (int r = 0; r < RV; r++) acc[r] -= 2.0f * dot;  // +qnorm@emit
       } else {
         const __half* Tm = Tq;
         for (int32_t m4 = 0; m4 < M / 4; m4++) {
 #pragma unroll
-          for (int r = 0; r < RPV; r++) {
+          for (int r = 0; r < RV; r++) {
             const uint32_t cw = code4[r][m4];
             acc[r] += __half2float(Tm[cw & 255]) +
                       __half2float(Tm[256 + ((cw >> 8) & 255)]) +
@@ -894,11 +900,11 @@ __global__ void __launch_bounds__(256, MINB) k_ivfpq_scan(
           Tm += 1024;
         }
 #pragma unroll
-        for (int r = 0; r < RPV; r++) acc[r] = -(acc[r] + dot);
+        for (int r = 0; r < RV; r++) acc[r] = -(acc[r] + dot);
       }
 #pragma unroll
-      for (int r = 0; r < RPV; r++) {
-        const int32_t v = lane + r * WAVE;
+      for (int r = 0; r < RV; r++) {
+        const int32_t v = vbase + lane + r * WAVE;
         if (v < tn) {
           const int64_t rw = t0 + v;
           bool pass = true;
@@ -2031,7 +2037,7 @@ void build_pass_bitmap(hipStream_t s, const int64_t* ids, int64_t n,
 // gfx950).  Blocks shrink below a wave for the rare huge-k calls — correct,
 // slow, and documented; host validates k <= 2048.
 static inline int select_threads(int32_t k) {
-  int T = (k <= 32) ? 256 : 64;
+  int T = 256;  // largest block whose per-thread lists fit 128 KB LDS
   while (T > 2 && 2ull * T * k * 8 > (128u << 10)) T >>= 1;
   return T;
 }
@@ -2329,9 +2335,17 @@ void ivfpq_scan(hipStream_t s, const uint32_t* units, int32_t n_units,
   // DG_PQ_RPV forces a tile for experiments).  TILE = RPV * 64 rows.
   const char* e = getenv("DG_PQ_RPV");
   int rpv = e ? atoi(e) : 4;  // 256-row tiles + 2 blocks/CU measured best
-  if (rpv != 4 && rpv != 6 && rpv != 8 && rpv != 16) rpv = 4;
-  size_t lds = (size_t)(rpv == 6 ? 4 : rpv) * 64 * M +
+  if (rpv != 4 && rpv != 6 && rpv != 8 && rpv != 16 && rpv != 17) rpv = 4;
+  size_t lds = (size_t)(rpv == 6 ? 4 : (rpv == 17 ? 16 : rpv)) * 64 * M +
                (size_t)M * 256 * 2;  // codes + f16 S_l
+  if (rpv == 17) {  // cooperative: 4 waves share each query's T window
+    hipLaunchKernelGGL((k_ivfpq_scan<16, 1, true>), dim3((uint32_t)n_units),
+                       dim3(256), lds, s, units, csr_offsets, csr_codes, S,
+                       T, coarse_dots, nlist, M, inv_offsets, inv_q,
+                       inv_rank, qp_off, q_cand_base, nprobe, metric, bitmap,
+                       chunk_rows, cand);
+    return;
+  }
   if (rpv == 6)  // round-1 form: VGPRs capped at 80 by launch_bounds(256,6)
     hipLaunchKernelGGL((k_ivfpq_scan<4, 6>), dim3((uint32_t)n_units),
                        dim3(256), lds, s, units, csr_offsets, csr_codes, S,
