@@ -1062,6 +1062,110 @@ __device__ __forceinline__ void dg_scan_rows_body(
   }
 }
 
+// v6 (variant 19): two banks x 8 dims, ordered wait(A) -> issue(B_next)
+// -> compute(A): the other bank's loads are issued a FULL 8-dim compute
+// (~1000 cy) before their drain, so each drain should find them landed.
+// Same full-drain tie discipline as the shipped pipeline.
+template <int JT>
+__device__ __forceinline__ void dg_scan_rows_body2(
+    const float* smem, const int64_t* cbase, const float* col,
+    const float* __restrict__ vnorms, int32_t d, int32_t nrows,
+    int32_t nrows_pad, int32_t rr0, int64_t row0, int32_t qt, int metric,
+    const uint32_t* __restrict__ bitmap, uint64_t* __restrict__ cand) {
+  constexpr int RPL = 4;
+  float acc[JT][RPL];
+#pragma unroll
+  for (int j = 0; j < JT; j++)
+#pragma unroll
+    for (int x = 0; x < RPL; x++) acc[j][x] = 0.f;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  const char* colp = (const char*)(col + rr0);
+  const size_t cstride = (size_t)nrows_pad * 4;
+  dg_f4 A[8], B[8];
+  auto issue8 = [&](dg_f4 (&b)[8], int32_t ib) {
+#pragma unroll
+    for (int u = 0; u < 8; u++)
+      DG_GLOAD4(b[u], colp + (size_t)(ib + u) * cstride);
+  };
+  auto wait8 = [&](dg_f4 (&b)[8]) {
+    asm volatile("s_waitcnt vmcnt(0)"
+                 : "+v"(b[0]), "+v"(b[1]), "+v"(b[2]), "+v"(b[3]),
+                   "+v"(b[4]), "+v"(b[5]), "+v"(b[6]), "+v"(b[7]));
+  };
+  auto compute8 = [&](dg_f4 (&b)[8], int32_t ib) {
+#pragma unroll
+    for (int j = 0; j < JT; j++) {
+      const float* qp = smem + (size_t)j * d + ib;
+#pragma unroll
+      for (int h = 0; h < 4; h++) {
+        const float2 qq = *(const float2*)(qp + 2 * h);
+#pragma unroll
+        for (int x = 0; x < RPL; x++)
+          acc[j][x] += b[2 * h][x] * qq.x + b[2 * h + 1][x] * qq.y;
+      }
+    }
+  };
+  int32_t base = 0;
+  if (d >= 16) {
+    issue8(A, 0);
+    for (; base + 16 <= d; base += 16) {
+      wait8(A);
+      if (base + 24 <= d) issue8(B, base + 8);
+      compute8(A, base);
+      if (base + 24 <= d) {
+        wait8(B);
+        if (base + 32 <= d) issue8(A, base + 16);
+        compute8(B, base + 8);
+      } else {
+        // last 8 dims of the main region handled by the tail below
+        base += 8;
+        break;
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  const float* colf = col + rr0;
+  for (int32_t ib = base; ib < d; ib += 4) {
+    float4 c[4];
+#pragma unroll
+    for (int u = 0; u < 4; u++)
+      c[u] = (ib + u < d)
+                 ? *(const float4*)(colf + (size_t)(ib + u) * nrows_pad)
+                 : float4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int j = 0; j < JT; j++) {
+      const float* qp = smem + (size_t)j * d + ib;
+#pragma unroll
+      for (int u = 0; u < 4 && ib + u < d; u++) {
+        const float qv = qp[u];
+        acc[j][0] += c[u].x * qv;
+        acc[j][1] += c[u].y * qv;
+        acc[j][2] += c[u].z * qv;
+        acc[j][3] += c[u].w * qv;
+      }
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#pragma unroll
+  for (int j = 0; j < JT; j++) {
+    if (j < qt) {
+      const int64_t cb = cbase[j] + row0 + rr0;
+#pragma unroll
+      for (int x = 0; x < RPL; x++) {
+        const int32_t rl = rr0 + x;
+        if (rl < nrows) {
+          const int64_t r = row0 + rl;
+          bool pass = true;
+          if (bitmap) pass = (bitmap[r >> 5] >> (r & 31)) & 1;
+          float key = (metric == 0) ? vnorms[r] - 2.0f * acc[j][x]
+                                    : -acc[j][x];
+          cand[cb + x] = pass ? pack_cand(key, (uint32_t)r) : kCandEmpty;
+        }
+      }
+    }
+  }
+}
+
 template <int QTM, bool SAFE = false>
 __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe(
     const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
@@ -1288,6 +1392,75 @@ __global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe3(
           }
         }
       }
+    }
+  }
+}
+
+
+template <int QTM>
+__global__ void __launch_bounds__(256, 1) k_ivf_scan_pipe4(
+    const uint32_t* __restrict__ units, const int64_t* __restrict__ csr_offsets,
+    const int32_t* __restrict__ chunk_off, const int64_t* __restrict__ chunk_base,
+    const float* __restrict__ tvec, const float* __restrict__ vnorms,
+    const float* __restrict__ queries, int32_t d,
+    const int32_t* __restrict__ inv_offsets, const int32_t* __restrict__ inv_q,
+    const int32_t* __restrict__ inv_rank, const int64_t* __restrict__ qp_off,
+    const int64_t* __restrict__ q_cand_base, int32_t nprobe, int metric,
+    const uint32_t* __restrict__ bitmap, int32_t chunk_rows,
+    uint64_t* __restrict__ cand) {
+  constexpr int RPL = 4;
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  int64_t* cbase = (int64_t*)(smem + (size_t)QTM * d);
+  const uint32_t list = units[2 * blockIdx.x];
+  const uint32_t chunk = units[2 * blockIdx.x + 1];
+  const int64_t list_start = csr_offsets[list];
+  const int64_t len = csr_offsets[list + 1] - list_start;
+  const int32_t nrows =
+      (int32_t)min((int64_t)chunk_rows, len - (int64_t)chunk * chunk_rows);
+  const int32_t nrows_pad = (nrows + 3) & ~3;
+  const float* col = tvec + chunk_base[chunk_off[list] + (int32_t)chunk];
+  const int64_t row0 = list_start + (int64_t)chunk * chunk_rows;
+  const int32_t iq0 = inv_offsets[list];
+  const int32_t nql = inv_offsets[list + 1] - iq0;
+  const int wave_id = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  __builtin_assume(d % 4 == 0 && d > 0);
+  for (int32_t t0 = 0; t0 < nql; t0 += QTM) {
+    const int32_t qt = min(QTM, nql - t0);
+    __syncthreads();
+    for (int32_t j = 0; j < qt; j++) {
+      int32_t q = inv_q[iq0 + t0 + j];
+      const float4* src = (const float4*)(queries + (size_t)q * d);
+      float4* dst = (float4*)(smem + (size_t)j * d);
+      for (int i = threadIdx.x; i < d / 4; i += blockDim.x) dst[i] = src[i];
+    }
+    for (size_t i = (size_t)qt * d + threadIdx.x; i < (size_t)QTM * d;
+         i += blockDim.x)
+      smem[i] = 0.f;
+    if (threadIdx.x < QTM) {
+      int32_t j = threadIdx.x;
+      if (j < qt) {
+        int32_t q = inv_q[iq0 + t0 + j];
+        int32_t rank = inv_rank[iq0 + t0 + j];
+        cbase[j] = q_cand_base[q] + qp_off[(int64_t)q * nprobe + rank] -
+                   list_start;
+      } else {
+        cbase[j] = 0;
+      }
+    }
+    __syncthreads();
+    for (int32_t rb = wave_id * WAVE * RPL; rb < nrows_pad;
+         rb += 4 * WAVE * RPL) {
+      const int32_t rr0 = rb + lane * RPL;
+      if (rr0 >= nrows_pad) continue;
+      if (QTM >= 16 && qt <= QTM / 2)
+        dg_scan_rows_body2<(QTM >= 16 ? QTM / 2 : QTM)>(
+            smem, cbase, col, vnorms, d, nrows, nrows_pad, rr0, row0, qt,
+            metric, bitmap, cand);
+      else
+        dg_scan_rows_body2<QTM>(smem, cbase, col, vnorms, d, nrows,
+                                nrows_pad, rr0, row0, qt, metric, bitmap,
+                                cand);
     }
   }
 }
@@ -2224,6 +2397,9 @@ void ivf_scan_col(hipStream_t s, const uint32_t* units, int32_t n_units,
 #undef DG_PIPE_LAUNCH
       break;
     }
+    // (case 19 pipe4 removed: 2x8-bank wait/issue/compute order also
+    // fails the hazard scan — 8 pending loads across a full compute push
+    // the allocator into stashing in-flight destinations)
     case 15: {  // round-1 columnar pipeline (previous default)
       DG_SCAN_LAUNCH(16, 4);
       break;
