@@ -11,6 +11,7 @@
 #include <cstring>
 
 #include <algorithm>
+#include <queue>
 #include <unordered_set>
 
 namespace dingogpu {
@@ -353,6 +354,92 @@ class GpuIvfPqIndex : public GpuIndexBase {
 
 }  // namespace
 
+
+// ---- reader brute-force fallback (vector_reader.cc:1873-2048) ----
+namespace {
+struct HeapEntry {  // DistanceResult analog (vector_reader.cc:1845-1858)
+  float distance;
+  VectorWithDistance vd;
+  bool operator<(const HeapEntry& o) const { return distance < o.distance; }
+};
+}  // namespace
+
+Status BruteForceSearch(MetricType metric, int32_t dimension,
+                        const RowIterator& next,
+                        const std::vector<VectorWithId>& queries,
+                        uint32_t topk,
+                        const std::vector<std::shared_ptr<FilterFunctor>>& f,
+                        const VectorSearchParameter& p,
+                        std::vector<VectorWithDistanceResult>& results,
+                        int64_t batch_count) {
+  if (dimension <= 0) return {kEVectorInvalid, "dimension invalid"};
+  std::vector<std::priority_queue<HeapEntry>> tops(queries.size());
+  std::vector<VectorWithId> batch;
+  batch.reserve(batch_count);
+  auto flush = [&]() -> Status {
+    if (batch.empty()) return Status::OK();
+    // throwaway per-batch Flat index (vector_reader.cc:1939-1942)
+    auto flat = NewFlatIndex(metric, dimension);
+    if (!flat) return {kEInternal, "flat index create failed"};
+    Status s = flat->Add(batch);
+    if (!s.ok()) return s;
+    std::vector<VectorWithDistanceResult> rb;
+    s = flat->Search(queries, topk, f, false, p, rb);
+    if (!s.ok()) return s;
+    for (size_t i = 0; i < rb.size(); i++) {
+      auto& top = tops[i];
+      for (auto& vd : rb[i].vector_with_distances) {
+        if (top.size() < topk) {
+          top.push({vd.distance, vd});
+        } else if (top.top().distance > vd.distance) {
+          top.pop();
+          top.push({vd.distance, vd});
+        }
+      }
+    }
+    batch.clear();
+    return Status::OK();
+  };
+  VectorWithId row;
+  while (next(&row)) {
+    batch.push_back(std::move(row));
+    if ((int64_t)batch.size() == batch_count) {
+      Status s = flush();
+      if (!s.ok()) return s;
+    }
+  }
+  Status s = flush();
+  if (!s.ok()) return s;
+  // ascending by distance (deque emplace_front, vector_reader.cc:2032-2044)
+  results.clear();
+  results.resize(queries.size());
+  for (size_t i = 0; i < tops.size(); i++) {
+    auto& top = tops[i];
+    std::vector<VectorWithDistance> tmp;
+    while (!top.empty()) {
+      tmp.push_back(top.top().vd);
+      top.pop();
+    }
+    results[i].vector_with_distances.assign(tmp.rbegin(), tmp.rend());
+  }
+  return Status::OK();
+}
+
+Status SearchWithBruteForceFallback(
+    VectorIndex* index, const std::function<RowIterator()>& scan_factory,
+    const std::vector<VectorWithId>& queries, uint32_t topk,
+    const std::vector<std::shared_ptr<FilterFunctor>>& f,
+    const VectorSearchParameter& p,
+    std::vector<VectorWithDistanceResult>& results) {
+  Status s = index->Search(queries, topk, f, false, p, results);
+  if (s.code == kEVectorNotSupport) {
+    // the reader's fallback trigger (vector_reader.cc:1828-1831)
+    return BruteForceSearch(index->GetMetricType(), index->GetDimension(),
+                            scan_factory(), queries, topk, f, p, results);
+  }
+  return s;
+}
+
 std::unique_ptr<VectorIndex> NewFlatIndex(MetricType metric, int32_t dim,
                                           int device) {
   auto p = std::make_unique<GpuFlatIndex>(metric, dim, device);
@@ -558,6 +645,80 @@ extern "C" int dg_mirror_selftest(void) {
         if (id == 3 || id == 9 || id == 15) return 56;
       }
     }
+  }
+
+  // ---- reader brute-force path + EVECTOR_NOT_SUPPORT fallback round trip
+  // (vector_reader.cc:1873-2048, :1828-1831) ----
+  DG_ST("brute force");
+  {
+    const int32_t d3 = 24;
+    const int64_t n3 = 5000;  // > 2 batches of 2048
+    std::vector<VectorWithId> rows(n3);
+    uint32_t s = 777;
+    for (int64_t i = 0; i < n3; i++) {
+      rows[i].id = i * 2 + 1;
+      rows[i].vector.dimension = d3;
+      rows[i].vector.float_values.resize(d3);
+      for (int j = 0; j < d3; j++) {
+        s = s * 1664525u + 1013904223u;
+        rows[i].vector.float_values[j] = (s >> 8) * (1.0f / 16777216.0f);
+      }
+    }
+    std::vector<VectorWithId> queries(rows.begin(), rows.begin() + 4);
+    auto make_scan = [&]() -> RowIterator {
+      auto pos = std::make_shared<int64_t>(0);
+      return [&rows, pos](VectorWithId* out) {
+        if (*pos >= (int64_t)rows.size()) return false;
+        *out = rows[(*pos)++];
+        return true;
+      };
+    };
+    VectorSearchParameter p3;
+    // ground truth: one persistent Flat index over all rows
+    auto all = NewFlatIndex(MetricType::kL2, d3);
+    if (!all) return 70;
+    if (!all->Add(rows).ok()) return 71;
+    std::vector<VectorWithDistanceResult> want;
+    if (!all->Search(queries, 7, {}, false, p3, want).ok()) return 72;
+    // KV-scan brute force (2048-row batches + heap merge) must agree
+    std::vector<VectorWithDistanceResult> got;
+    if (!BruteForceSearch(MetricType::kL2, d3, make_scan(), queries, 7, {},
+                          p3, got).ok())
+      return 73;
+    if (got.size() != want.size()) return 74;
+    for (size_t i = 0; i < got.size(); i++) {
+      auto& g = got[i].vector_with_distances;
+      auto& w = want[i].vector_with_distances;
+      if (g.size() != w.size()) return 75;
+      for (size_t j = 0; j < g.size(); j++) {
+        if (g[j].vector_with_id.id != w[j].vector_with_id.id) return 76;
+        if (std::fabs(g[j].distance - w[j].distance) > 1e-4f) return 77;
+      }
+    }
+    // NOT_SUPPORT fallback round trip: an index that rejects the request
+    // must transparently drop to the brute-force scan
+    class NotSupportIndex : public GpuFlatIndex {
+     public:
+      NotSupportIndex(MetricType m, int32_t d) : GpuFlatIndex(m, d, -1) {}
+      Status Search(const std::vector<VectorWithId>&, uint32_t,
+                    const std::vector<std::shared_ptr<FilterFunctor>>&,
+                    bool, const VectorSearchParameter&,
+                    std::vector<VectorWithDistanceResult>&) override {
+        return {kEVectorNotSupport, "not support"};
+      }
+    };
+    NotSupportIndex ns(MetricType::kL2, d3);
+    std::vector<VectorWithDistanceResult> got2;
+    if (!SearchWithBruteForceFallback(&ns, make_scan, queries, 7, {}, p3,
+                                      got2).ok())
+      return 78;
+    if (got2.size() != want.size()) return 79;
+    for (size_t i = 0; i < got2.size(); i++)
+      if (got2[i].vector_with_distances.size() !=
+              want[i].vector_with_distances.size() ||
+          got2[i].vector_with_distances[0].vector_with_id.id !=
+              want[i].vector_with_distances[0].vector_with_id.id)
+        return 80;
   }
   DG_ST("done");
   return 0;

@@ -10,6 +10,7 @@
 #pragma once
 
 #include <cstdint>
+#include <functional>
 #include <memory>
 #include <string>
 #include <vector>
@@ -182,6 +183,35 @@ class VectorIndex {
       const VectorSearchParameter& p,
       std::vector<VectorWithDistanceResult>& results) = 0;
 };
+
+// ---- reader brute-force fallback ----
+// Mirror of VectorReader::BruteForceSearch (vector_reader.cc:1873-2048):
+// scan the region's rows in FLAGS_vector_index_bruteforce_batch_count
+// batches (default 2048, vector_reader.cc:61), build a throwaway Flat
+// index per batch, search it, and merge per-query top-k by the dingo
+// distance (max-heap keeps the smallest topk; output ascending).  The KV
+// iterator is abstracted as a pull function (the real reader wires the
+// RocksDB range scan + proto decode here, vector_reader.cc:1922-1936).
+using RowIterator = std::function<bool(VectorWithId*)>;
+
+Status BruteForceSearch(MetricType metric, int32_t dimension,
+                        const RowIterator& next,
+                        const std::vector<VectorWithId>& queries,
+                        uint32_t topk,
+                        const std::vector<std::shared_ptr<FilterFunctor>>& f,
+                        const VectorSearchParameter& p,
+                        std::vector<VectorWithDistanceResult>& results,
+                        int64_t batch_count = 2048);
+
+// Search with the reader's EVECTOR_NOT_SUPPORT fallback round trip
+// (vector_reader.cc:1828-1831): an index that rejects the request drops
+// to the KV-scan brute force.  scan_factory yields a fresh iterator.
+Status SearchWithBruteForceFallback(
+    VectorIndex* index, const std::function<RowIterator()>& scan_factory,
+    const std::vector<VectorWithId>& queries, uint32_t topk,
+    const std::vector<std::shared_ptr<FilterFunctor>>& f,
+    const VectorSearchParameter& p,
+    std::vector<VectorWithDistanceResult>& results);
 
 // GPU-backed concrete indexes (Flat / IVF-Flat), the factory, and a
 // self-test used by the GPU test suite.
