@@ -1959,8 +1959,14 @@ __global__ void __launch_bounds__(256, 2) k_dots_mfma(
   // 128x128x32 f32 shape reaches 122 TF; a 64x64/BK16 first cut measured
   // 27 TF — barrier overhead per 8 MFMAs dominated).
   constexpr int BM = 128, BN = 128, BK = 32;
-  __shared__ float lx[2][BM][BK + 1];  // odd stride: conflict-free column
-  __shared__ float ly[2][BN][BK + 1];  //   reads; scalar b32 staging writes
+  // BK+4 stride: rows are 144 B = 16 B-aligned, so staging writes are
+  // ds_write_b128 (the round-1 BK+1 odd stride forced 32 SCALAR b32
+  // writes per thread per stage — the staging cost, not the MFMAs, capped
+  // that kernel at 57 TF).  Column reads pay a 2-way bank conflict
+  // (rows r and r+16 share banks at stride 36) — measured cheaper than
+  // the scalar-write staging.
+  __shared__ float lx[2][BM][BK + 4];
+  __shared__ float ly[2][BN][BK + 4];
   using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
 
   const int64_t m0 = (int64_t)blockIdx.y * BM;
@@ -1984,14 +1990,8 @@ __global__ void __launch_bounds__(256, 2) k_dots_mfma(
       for (int t = 0; t < 4; t++) {
         const float4 vx = *(const float4*)&X[xm * K + k0 + sk + 4 * t];
         const float4 vy = *(const float4*)&Y[yn * K + k0 + sk + 4 * t];
-        lx[buf][sr][sk + 4 * t + 0] = vx.x;
-        lx[buf][sr][sk + 4 * t + 1] = vx.y;
-        lx[buf][sr][sk + 4 * t + 2] = vx.z;
-        lx[buf][sr][sk + 4 * t + 3] = vx.w;
-        ly[buf][sr][sk + 4 * t + 0] = vy.x;
-        ly[buf][sr][sk + 4 * t + 1] = vy.y;
-        ly[buf][sr][sk + 4 * t + 2] = vy.z;
-        ly[buf][sr][sk + 4 * t + 3] = vy.w;
+        *(float4*)&lx[buf][sr][sk + 4 * t] = vx;
+        *(float4*)&ly[buf][sr][sk + 4 * t] = vy;
       }
     } else {  // edge tile: element-wise with zero padding
 #pragma unroll
