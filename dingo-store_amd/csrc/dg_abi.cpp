@@ -1964,8 +1964,26 @@ extern "C" dg_status dg_save(dg_index* ix, const char* path) {
   return st;
 }
 
+static dg_status dg_load_impl(dg_index** out, const char* path,
+                              int32_t device);
+
 extern "C" dg_status dg_load(dg_index** out, const char* path,
                              int32_t device) {
+  // exception wall: a corrupt container header can drive the staging
+  // vector resizes into std::bad_alloc; nothing may cross the C ABI
+  try {
+    return dg_load_impl(out, path, device);
+  } catch (const std::exception& e) {
+    dg_set_error("container load failed: %s", e.what());
+    return DG_EIO;
+  } catch (...) {
+    dg_set_error("container load failed");
+    return DG_EIO;
+  }
+}
+
+static dg_status dg_load_impl(dg_index** out, const char* path,
+                              int32_t device) {
   if (!out || !path) return DG_EINVAL;
   FILE* f = fopen(path, "rb");
   if (!f) {

@@ -31,6 +31,7 @@
 // like the reference (vector_index_flat.cc:88-91).
 #include <cstdio>
 #include <cstring>
+#include <stdexcept>
 #include <vector>
 
 #include "dg_internal.h"
@@ -371,8 +372,26 @@ extern "C" dg_status dg_save_faiss(dg_index* ix, const char* path) {
   return DG_OK;
 }
 
+static dg_status load_faiss_impl(dg_index** out, const char* path,
+                                 int32_t metric_override, int32_t device);
+
 extern "C" dg_status dg_load_faiss(dg_index** out, const char* path,
                                    int32_t metric_override, int32_t device) {
+  // exception wall: a malformed container can drive a vector resize into
+  // std::bad_alloc; no exception may cross the C ABI
+  try {
+    return load_faiss_impl(out, path, metric_override, device);
+  } catch (const std::exception& e) {
+    dg_set_error("faiss container load failed: %s", e.what());
+    return DG_EIO;
+  } catch (...) {
+    dg_set_error("faiss container load failed");
+    return DG_EIO;
+  }
+}
+
+static dg_status load_faiss_impl(dg_index** out, const char* path,
+                                 int32_t metric_override, int32_t device) {
   if (!out || !path) return DG_EINVAL;
   FILE* f = fopen(path, "rb");
   if (!f) {
@@ -443,8 +462,14 @@ extern "C" dg_status dg_load_faiss(dg_index** out, const char* path,
       int32_t M = 0;
       if (h4 == fcc("IwPQ")) {
         desc.kind = DG_INDEX_IVF_PQ;
-        (void)r.u8();  // by_residual (faiss default true; we require it)
+        uint8_t by_residual = r.u8();
         uint64_t code_size = r.u64();
+        if (!by_residual) {
+          // non-residual IVFPQ files would decode with the wrong ADC
+          dg_set_error("IwPQ with by_residual=0 unsupported");
+          st = DG_ENOT_SUPPORT;
+          break;
+        }
         uint64_t pq_d = r.u64();
         uint64_t pq_M = r.u64();
         uint64_t pq_nbits = r.u64();
