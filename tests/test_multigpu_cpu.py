@@ -72,17 +72,18 @@ def _worker(rank, world, port, ret):
     dist.destroy_process_group()
 
 
-def test_row_shard_merge_gloo_ws2():
+@pytest.mark.parametrize("world,port", [(2, 29781), (3, 29782)])
+def test_row_shard_merge_gloo(world, port):
+    # world=3 exercises uneven row shards (8000 % 3 != 0)
     ctx = mp.get_context("spawn")
     mgr = ctx.Manager()
     ret = mgr.dict()
-    port = 29781
-    ps = [ctx.Process(target=_worker, args=(r, 2, port, ret))
-          for r in range(2)]
+    ps = [ctx.Process(target=_worker, args=(r, world, port, ret))
+          for r in range(world)]
     for p in ps:
         p.start()
     for p in ps:
-        p.join(timeout=120)
+        p.join(timeout=180)
         assert p.exitcode == 0
     assert ret["ids_equal"]
     assert ret["dist_close"]
