@@ -14,8 +14,7 @@ for path in sys.argv[1:-1]:
     q = f"""
     SELECT ks.display_name, pi.name, SUM(pe.value), COUNT(*)
     FROM rocpd_pmc_event_{sfx} pe
-    JOIN rocpd_kernel_dispatch_{sfx} kd
-      ON pe.event_id = kd.event_id OR pe.event_id = kd.id
+    JOIN rocpd_kernel_dispatch_{sfx} kd ON pe.event_id = kd.event_id
     JOIN rocpd_info_kernel_symbol_{sfx} ks ON kd.kernel_id = ks.id
     JOIN rocpd_info_pmc_{sfx} pi ON pe.pmc_id = pi.id
     GROUP BY ks.display_name, pi.name"""
