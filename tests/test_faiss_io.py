@@ -54,6 +54,30 @@ def test_golden_container_structure():
     assert idx["direct_map"]["type"] == 0 and len(idx["direct_map"]["array"]) == 0
 
 
+def test_golden_container_searches_like_raw_vectors():
+    """Semantic pin, CPU-only: an exact search over the vectors as stored
+    IN the container must equal the exact search over the original raw
+    vectors — i.e. dg_save_faiss wrote the right bytes, not just the right
+    structure.  Uses the oracle as the checker."""
+    if not os.path.exists(GOLDEN):
+        pytest.skip("golden faiss container not generated yet")
+    import pyoracle as orc
+    exp = np.load(GOLDEN_NPZ)
+    idx = ff.read_index(GOLDEN)
+    # flatten container rows (vectors + their stored ids)
+    vecs = np.concatenate(
+        [c.view(np.float32).reshape(-1, idx["header"]["d"])
+         if len(c) else np.empty((0, idx["header"]["d"]), np.float32)
+         for c in idx["invlists"]["codes"]])
+    ids = np.concatenate(idx["invlists"]["ids"]).astype(np.int64)
+    q = workload.gen_queries(404, len(exp["ids"]), int(exp["d"]), 16)
+    cd, ci = orc.flat_search(orc.L2, vecs, q, 5, ids=ids)
+    rd, ri = orc.flat_search(orc.L2, np.asarray(exp["vectors"]), q, 5,
+                             ids=np.asarray(exp["ids"], np.int64))
+    np.testing.assert_array_equal(ci, ri)
+    np.testing.assert_allclose(cd, rd, rtol=1e-6, atol=1e-6)
+
+
 # ---------------- GPU round trips ----------------
 gpu = pytest.mark.gpu
 
