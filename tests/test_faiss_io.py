@@ -119,12 +119,14 @@ def test_flat_roundtrip(tmp_path, metric_name):
 
 
 @gpu
-def test_ivfflat_roundtrip(tmp_path):
+@pytest.mark.parametrize("metric_name", ["l2", "ip", "cosine"])
+def test_ivfflat_roundtrip(tmp_path, metric_name):
     dg = _dg()
+    metric = {"l2": dg.L2, "ip": dg.IP, "cosine": dg.COSINE}[metric_name]
     n, d, nq, k, nlist, nprobe = 30000, 96, 64, 10, 64, 16
     base = workload.gen_base(202, n, d)
     q = workload.gen_queries(202, n, d, nq)
-    idx = dg.Index(dg.IVF_FLAT, dg.L2, d, nlist=nlist)
+    idx = dg.Index(dg.IVF_FLAT, metric, d, nlist=nlist)
     idx.train(base[:8000])
     ids = np.arange(n, dtype=np.int64)
     idx.add(ids, base)
@@ -135,8 +137,10 @@ def test_ivfflat_roundtrip(tmp_path):
     parsed = ff.read_index(p)
     assert parsed["kind"] == "ivfflat" and parsed["nlist"] == nlist
     assert int(sum(parsed["invlists"]["sizes"])) == n
+    # cosine is stored as IP over normalized vectors (flat.cc:88-91)
+    assert parsed["header"]["metric"] == (1 if metric == dg.L2 else 0)
 
-    idx2 = dg.Index.load_faiss(p)
+    idx2 = dg.Index.load_faiss(p, metric=metric)
     d1, i1 = idx2.search(q, k, nprobe=nprobe)
     np.testing.assert_array_equal(i0, i1)
     np.testing.assert_allclose(d0, d1, rtol=1e-6, atol=1e-6)
