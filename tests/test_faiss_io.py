@@ -229,3 +229,18 @@ def test_golden_generation_and_load(tmp_path):
         ga.close()
         gb.close()
     idx.close()
+
+
+def test_parser_rejects_malformed():
+    """The independent parser must fail loudly on truncated or corrupt
+    containers (and so must dg_load_faiss, via its exception wall — the
+    GPU round-trip tests cover that side)."""
+    if not os.path.exists(GOLDEN):
+        pytest.skip("golden faiss container not generated yet")
+    data = open(GOLDEN, "rb").read()
+    with pytest.raises(Exception):
+        ff.read_index(data[: len(data) // 2])  # truncated mid-invlists
+    with pytest.raises(Exception):
+        ff.read_index(b"\x00\x01\x02\x03" + data[4:])  # bad fourcc
+    with pytest.raises(Exception):
+        ff.read_index(data + b"junk")  # trailing bytes
