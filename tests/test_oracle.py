@@ -308,3 +308,37 @@ def test_golden_pq_and_range():
     assert np.array_equal(rl, g["range_lims"])
     assert np.array_equal(ri, g["range_ids"])
     assert np.array_equal(rd, g["range_dist"])
+
+
+def test_bruteforce_batching_equals_whole_scan():
+    """Reader brute-force semantics (vector_reader.cc:1873-2048): scanning
+    in 2048-row batches through per-batch exact Flat searches and merging
+    per-query top-k by distance (max-heap keeps the smallest topk; output
+    ascending) must equal one exact search over all rows.  This is the CPU
+    statement of the property dg_mirror_selftest pins on the GPU mirror."""
+    import heapq
+    rng = np.random.default_rng(99)
+    n, d, nq, k, batch = 5000, 24, 6, 7, 2048
+    base = rng.random((n, d), dtype=np.float32)
+    ids = np.arange(n, dtype=np.int64) * 2 + 1
+    q = base[:nq] + rng.normal(0, 0.05, (nq, d)).astype(np.float32)
+
+    wd, wi = orc.flat_search(orc.L2, base, q, k, ids=ids)
+
+    tops = [[] for _ in range(nq)]  # python heapq = min-heap; negate
+    for s0 in range(0, n, batch):
+        bd, bi = orc.flat_search(orc.L2, base[s0:s0 + batch], q, k,
+                                 ids=ids[s0:s0 + batch])
+        for r in range(nq):
+            for dist, vid in zip(bd[r], bi[r]):
+                if vid < 0:
+                    continue
+                if len(tops[r]) < k:
+                    heapq.heappush(tops[r], (-dist, -vid))
+                elif -tops[r][0][0] > dist:  # current worst > new
+                    heapq.heapreplace(tops[r], (-dist, -vid))
+    for r in range(nq):
+        merged = sorted(((-md, -mv) for md, mv in tops[r]))
+        np.testing.assert_array_equal([v for _, v in merged], wi[r])
+        np.testing.assert_allclose([dd for dd, _ in merged], wd[r],
+                                   rtol=1e-6, atol=1e-6)
