@@ -10,4 +10,4 @@ done
 echo "=== VARIANT 10 batch 8192 ==="
 DG_SCAN_VARIANT=10 timeout 300 python bench.py --steps 5 --warmup 2 --batch 8192 --no-cpu-baseline --no-recall 2>bench_err.txt | python -c "import json,sys; j=json.load(sys.stdin); print('qps', j['value'], 'ms', j['ms_per_step'], 'scan_ms', j['roofline']['detail']['scan_ms_per_launch'], 'frac', j['roofline']['frac'])" || tail -12 bench_err.txt
 echo "=== PQ tile A/B (cfg D) ==="
-timeout 900 python pq_ab.py 2>pq_err.txt || tail -12 pq_err.txt
+timeout 900 python tools/pq_ab.py 2>pq_err.txt || tail -12 pq_err.txt
