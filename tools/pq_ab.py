@@ -1,6 +1,6 @@
 """A/B the IVF-PQ scan tile (DG_PQ_RPV) on one cfg D index build."""
 import argparse, os, sys, time
-REPO = os.path.dirname(os.path.abspath(__file__))
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path[:0] = [REPO, os.path.join(REPO, "dingo-store_amd")]
 import torch
 import bench as B
